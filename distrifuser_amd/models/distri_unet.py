@@ -1,0 +1,237 @@
+"""DistriUNet — model-level parallel orchestration around the native U-Net.
+
+Handles (semantics parity with the reference model wrappers,
+/root/reference/distrifuser/models/*.py, re-designed):
+
+* CFG batch split: the 2-sample guidance batch is split across the two rank
+  halves; each rank computes one branch.
+* patch parallelism: the U-Net itself consumes the full latent and emits this
+  rank's row band (conv_in slices); the bands are re-assembled with one
+  world all-gather per step.
+* naive patch parallelism: the latent is sliced with NO cross-patch
+  interaction (quality baseline), row/col/alternate schemes.
+* tensor parallelism: full latent per rank, CFG halves exchanged pairwise.
+* hipGraph capture (torch.cuda.CUDAGraph == hipGraph on ROCm) of the whole
+  step, including the RCCL collectives, keyed by warmup phase.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+from ..parallel.state import ParallelState
+from ..utils.comm import PatchParallelismCommManager
+from ..utils.config import DistriConfig
+from .unet import UNet2DConditionNative, UNetConfig
+
+
+class DistriUNet(nn.Module):
+    def __init__(self, unet_config: UNetConfig, distri_config: DistriConfig):
+        super().__init__()
+        self.distri_config = distri_config
+        self.state = ParallelState(distri_config)
+        self.unet = UNet2DConditionNative(unet_config, self.state)
+        self.comm_manager: PatchParallelismCommManager | None = None
+
+        self.buffer_list: list[torch.Tensor] | None = None
+        self.output_buffer: torch.Tensor | None = None
+
+        self.static_inputs: dict | None = None
+        self.static_outputs: list | None = None
+        self.cuda_graphs: list | None = None
+
+    @property
+    def config(self) -> UNetConfig:
+        return self.unet.config
+
+    # -- control plane (reference base_model.py:27-52) ----------------------
+
+    def set_counter(self, counter: int = 0) -> None:
+        self.state.set_counter(counter)
+
+    def set_comm_manager(self, comm_manager: PatchParallelismCommManager) -> None:
+        self.comm_manager = comm_manager
+        self.state.comm_manager = comm_manager
+
+    def reset_lazy_modules(self) -> None:
+        for m in self.unet.modules():
+            if hasattr(m, "reset") and m is not self:
+                m.reset()
+
+    def synchronize(self) -> None:
+        if self.comm_manager is not None and self.comm_manager.handles:
+            self.comm_manager.clear()
+
+    def setup_cuda_graph(self, static_outputs, cuda_graphs) -> None:
+        self.static_outputs = static_outputs
+        self.cuda_graphs = cuda_graphs
+
+    # -- helpers -------------------------------------------------------------
+
+    def _naive_slice(self, sample: torch.Tensor) -> tuple[torch.Tensor, int]:
+        cfg = self.distri_config
+        n = cfg.n_device_per_batch
+        scheme = cfg.split_scheme
+        if scheme == "alternate":
+            scheme = "row" if self.state.counter % 2 == 0 else "col"
+        dim = 2 if scheme == "row" else 3
+        size = sample.shape[dim]
+        assert size % n == 0
+        chunk = size // n
+        idx = cfg.split_idx()
+        sl = [slice(None)] * 4
+        sl[dim] = slice(idx * chunk, (idx + 1) * chunk)
+        return sample[tuple(sl)].contiguous(), dim
+
+    def _gather_patches(self, local: torch.Tensor, dim: int, b_full: int) -> torch.Tensor:
+        """All-gather per-rank patches over WORLD and reassemble (2,C,H,W)."""
+        cfg = self.distri_config
+        ws = cfg.world_size
+        n = cfg.n_device_per_batch
+        local = local.contiguous()
+        if self.buffer_list is None or self.buffer_list[0].shape != local.shape:
+            self.buffer_list = [torch.empty_like(local) for _ in range(ws)]
+            self.output_buffer = None
+        dist.all_gather(self.buffer_list, local, async_op=False)
+        if cfg.split_batch:
+            b0 = torch.cat(self.buffer_list[:n], dim=dim)
+            b1 = torch.cat(self.buffer_list[n:], dim=dim)
+            out = torch.cat([b0, b1], dim=0)
+        else:
+            out = torch.cat(self.buffer_list, dim=dim)
+        if self.output_buffer is None or self.output_buffer.shape != out.shape:
+            self.output_buffer = torch.empty_like(out)
+        self.output_buffer.copy_(out)
+        return self.output_buffer
+
+    def _slice_cfg(self, sample, timestep, encoder_hidden_states, added_cond_kwargs):
+        cfg = self.distri_config
+        bi = cfg.batch_idx()
+        sample = sample[bi : bi + 1]
+        if torch.is_tensor(timestep) and timestep.ndim > 0 and timestep.shape[0] > 1:
+            timestep = timestep[bi : bi + 1]
+        encoder_hidden_states = encoder_hidden_states[bi : bi + 1]
+        if added_cond_kwargs is not None:
+            added_cond_kwargs = {k: v[bi : bi + 1] for k, v in added_cond_kwargs.items()}
+        return sample, timestep, encoder_hidden_states, added_cond_kwargs
+
+    # -- forward -------------------------------------------------------------
+
+    def forward(
+        self,
+        sample: torch.Tensor,
+        timestep,
+        encoder_hidden_states: torch.Tensor,
+        added_cond_kwargs: dict | None = None,
+        record: bool = False,
+    ) -> torch.Tensor:
+        cfg = self.distri_config
+        state = self.state
+        b, c, h, w = sample.shape
+
+        graph_ready = (
+            self.cuda_graphs is not None
+            and self.static_inputs is not None
+            and not record
+        )
+        if graph_ready:
+            self._copy_static_inputs(sample, timestep, encoder_hidden_states, added_cond_kwargs)
+            if cfg.parallelism == "naive_patch" and cfg.split_scheme == "alternate":
+                gi = state.counter % len(self.cuda_graphs)
+            elif state.counter <= cfg.warmup_steps:
+                gi = 0
+            elif state.counter == cfg.warmup_steps + 1:
+                gi = min(1, len(self.cuda_graphs) - 1)
+            else:
+                gi = min(2, len(self.cuda_graphs) - 1)
+            self.cuda_graphs[gi].replay()
+            state.next_step()
+            return self.static_outputs[gi]
+
+        state.recording = record
+        try:
+            output = self._forward_impl(
+                sample, timestep, encoder_hidden_states, added_cond_kwargs, b, c, h, w
+            )
+        finally:
+            state.recording = False
+
+        if record and self.static_inputs is None:
+            self.static_inputs = {
+                "sample": sample,
+                "timestep": timestep,
+                "encoder_hidden_states": encoder_hidden_states,
+                "added_cond_kwargs": added_cond_kwargs,
+            }
+        if record:
+            self.synchronize()
+
+        state.next_step()
+        return output
+
+    def _forward_impl(self, sample, timestep, encoder_hidden_states, added_cond_kwargs, b, c, h, w):
+        cfg = self.distri_config
+        if cfg.world_size == 1:
+            return self.unet(sample, timestep, encoder_hidden_states, added_cond_kwargs)
+
+        if cfg.parallelism == "patch":
+            if cfg.split_batch:
+                assert b == 2, f"CFG split expects batch 2, got {b}"
+                sample, timestep, encoder_hidden_states, added_cond_kwargs = self._slice_cfg(
+                    sample, timestep, encoder_hidden_states, added_cond_kwargs
+                )
+            local = self.unet(sample, timestep, encoder_hidden_states, added_cond_kwargs)
+            if cfg.n_device_per_batch == 1 and not cfg.split_batch:
+                return local
+            return self._gather_patches(local, dim=2, b_full=b)
+
+        if cfg.parallelism == "naive_patch":
+            if cfg.split_batch:
+                assert b == 2
+                sample, timestep, encoder_hidden_states, added_cond_kwargs = self._slice_cfg(
+                    sample, timestep, encoder_hidden_states, added_cond_kwargs
+                )
+            patch, dim = self._naive_slice(sample)
+            local = self.unet(patch, timestep, encoder_hidden_states, added_cond_kwargs)
+            return self._gather_patches(local, dim=dim, b_full=b)
+
+        if cfg.parallelism == "tensor":
+            if cfg.split_batch:
+                assert b == 2
+                sample, timestep, encoder_hidden_states, added_cond_kwargs = self._slice_cfg(
+                    sample, timestep, encoder_hidden_states, added_cond_kwargs
+                )
+                local = self.unet(sample, timestep, encoder_hidden_states, added_cond_kwargs)
+                local = local.contiguous()
+                # exchange the two CFG halves of the same (full) latent
+                # (fixes the reference's broken split_group() call,
+                # distri_sdxl_unet_tp.py:159-162)
+                pair = [torch.empty_like(local) for _ in range(2)]
+                dist.all_gather(pair, local, group=cfg.split_group)
+                return torch.cat(pair, dim=0)
+            return self.unet(sample, timestep, encoder_hidden_states, added_cond_kwargs)
+
+        raise ValueError(f"unknown parallelism {cfg.parallelism!r}")
+
+    # -- hipGraph static-input plumbing --------------------------------------
+
+    def _copy_static_inputs(self, sample, timestep, encoder_hidden_states, added_cond_kwargs):
+        cfg = self.distri_config
+        si = self.static_inputs
+        if cfg.split_batch and sample.shape[0] == 2 and si["sample"].shape[0] == 1:
+            sample, timestep, encoder_hidden_states, added_cond_kwargs = self._slice_cfg(
+                sample, timestep, encoder_hidden_states, added_cond_kwargs
+            )
+        si["sample"].copy_(sample)
+        ts = si["timestep"]
+        if torch.is_tensor(ts):
+            if torch.is_tensor(timestep):
+                ts.copy_(timestep.reshape(ts.shape))
+            else:
+                ts.fill_(timestep)
+        si["encoder_hidden_states"].copy_(encoder_hidden_states)
+        if added_cond_kwargs is not None and si["added_cond_kwargs"] is not None:
+            for k in added_cond_kwargs:
+                si["added_cond_kwargs"][k].copy_(added_cond_kwargs[k])

@@ -1,0 +1,159 @@
+"""Native VAE (AutoencoderKL) decoder.
+
+The reference delegated VAE decode to diffusers; here it is owned. Every
+rank decodes the identical full latent (parity with the reference's
+replicated decode — the denoise loop is the parallel part). The mid-block
+attention is evaluated with a query-chunked online-softmax fallback so that
+3840x3840 (480x480 = 230k tokens, single 512-dim head) decodes without
+materializing the score matrix; on GPU it rides the fused SDPA.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from .. import ops
+
+
+@dataclass(frozen=True)
+class VAEDecoderConfig:
+    latent_channels: int = 4
+    out_channels: int = 3
+    block_out_channels: tuple = (128, 256, 512, 512)
+    layers_per_block: int = 2
+    norm_num_groups: int = 32
+    scaling_factor: float = 0.13025  # SDXL
+
+
+SDXL_VAE = VAEDecoderConfig()
+SD_VAE = VAEDecoderConfig(scaling_factor=0.18215)
+# 4 up blocks = 8x upscale, same ratio as the real VAE
+TINY_VAE = VAEDecoderConfig(
+    latent_channels=4, block_out_channels=(16, 16, 32, 32), layers_per_block=1,
+    norm_num_groups=8, scaling_factor=0.18215,
+)
+
+
+class VAEResnetBlock(nn.Module):
+    """ResNet block without time embedding (VAE variant)."""
+
+    def __init__(self, in_ch: int, out_ch: int, groups: int):
+        super().__init__()
+        self.norm1 = nn.GroupNorm(groups, in_ch, eps=1e-6)
+        self.conv1 = nn.Conv2d(in_ch, out_ch, 3, padding=1)
+        self.norm2 = nn.GroupNorm(groups, out_ch, eps=1e-6)
+        self.conv2 = nn.Conv2d(out_ch, out_ch, 3, padding=1)
+        self.conv_shortcut = nn.Conv2d(in_ch, out_ch, 1) if in_ch != out_ch else None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        h = self.conv1(ops.group_norm_silu(x, self.norm1.num_groups, self.norm1.weight,
+                                           self.norm1.bias, self.norm1.eps))
+        h = self.conv2(ops.group_norm_silu(h, self.norm2.num_groups, self.norm2.weight,
+                                           self.norm2.bias, self.norm2.eps))
+        if self.conv_shortcut is not None:
+            x = self.conv_shortcut(x)
+        return x + h
+
+
+def _chunked_single_head_attention(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, chunk: int = 16384
+) -> torch.Tensor:
+    """[B, L, C] single-head attention with O(chunk*L) memory."""
+    scale = q.shape[-1] ** -0.5
+    outs = []
+    for s in range(0, q.shape[1], chunk):
+        scores = torch.einsum("bqc,bkc->bqk", q[:, s : s + chunk].float() * scale, k.float())
+        outs.append(torch.einsum("bqk,bkc->bqc", scores.softmax(dim=-1), v.float()))
+    return torch.cat(outs, dim=1).to(q.dtype)
+
+
+class VAEAttention(nn.Module):
+    """Single-head spatial self-attention of the VAE mid block."""
+
+    def __init__(self, channels: int, groups: int):
+        super().__init__()
+        self.group_norm = nn.GroupNorm(groups, channels, eps=1e-6)
+        self.to_q = nn.Linear(channels, channels)
+        self.to_k = nn.Linear(channels, channels)
+        self.to_v = nn.Linear(channels, channels)
+        self.to_out = nn.Linear(channels, channels)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        b, c, h, w = x.shape
+        residual = x
+        x = ops.group_norm_silu(x, self.group_norm.num_groups, self.group_norm.weight,
+                                self.group_norm.bias, self.group_norm.eps, silu=False)
+        x = x.permute(0, 2, 3, 1).reshape(b, h * w, c)
+        q, k, v = self.to_q(x), self.to_k(x), self.to_v(x)
+        if x.is_cuda and c <= 256:
+            out = ops.flash_attention(q[:, None], k[:, None], v[:, None])[:, 0]
+        else:
+            out = _chunked_single_head_attention(q, k, v)
+        out = self.to_out(out)
+        return out.reshape(b, h, w, c).permute(0, 3, 1, 2) + residual
+
+
+class UpDecoderBlock2D(nn.Module):
+    def __init__(self, in_ch: int, out_ch: int, layers: int, add_upsample: bool, groups: int):
+        super().__init__()
+        self.resnets = nn.ModuleList(
+            [VAEResnetBlock(in_ch if i == 0 else out_ch, out_ch, groups) for i in range(layers)]
+        )
+        self.upsampler = nn.Conv2d(out_ch, out_ch, 3, padding=1) if add_upsample else None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        for r in self.resnets:
+            x = r(x)
+        if self.upsampler is not None:
+            x = F.interpolate(x, scale_factor=2.0, mode="nearest")
+            x = self.upsampler(x)
+        return x
+
+
+class VAEDecoder(nn.Module):
+    def __init__(self, config: VAEDecoderConfig):
+        super().__init__()
+        self.config = config
+        ch = config.block_out_channels
+        groups = config.norm_num_groups
+        top = ch[-1]
+        self.post_quant_conv = nn.Conv2d(config.latent_channels, config.latent_channels, 1)
+        self.conv_in = nn.Conv2d(config.latent_channels, top, 3, padding=1)
+        self.mid_resnet_1 = VAEResnetBlock(top, top, groups)
+        self.mid_attn = VAEAttention(top, groups)
+        self.mid_resnet_2 = VAEResnetBlock(top, top, groups)
+        rev = list(reversed(ch))
+        blocks = []
+        prev = top
+        for i, out_ch in enumerate(rev):
+            blocks.append(
+                UpDecoderBlock2D(
+                    prev, out_ch, config.layers_per_block + 1,
+                    add_upsample=i < len(rev) - 1, groups=groups,
+                )
+            )
+            prev = out_ch
+        self.up_blocks = nn.ModuleList(blocks)
+        self.conv_norm_out = nn.GroupNorm(groups, ch[0], eps=1e-6)
+        self.conv_out = nn.Conv2d(ch[0], config.out_channels, 3, padding=1)
+
+    @torch.no_grad()
+    def decode(self, latents: torch.Tensor) -> torch.Tensor:
+        """latents (scaled) -> images in [-1, 1]."""
+        z = latents / self.config.scaling_factor
+        z = self.post_quant_conv(z)
+        x = self.conv_in(z)
+        x = self.mid_resnet_1(x)
+        x = self.mid_attn(x)
+        x = self.mid_resnet_2(x)
+        for block in self.up_blocks:
+            x = block(x)
+        x = ops.group_norm_silu(x, self.conv_norm_out.num_groups, self.conv_norm_out.weight,
+                                self.conv_norm_out.bias, self.conv_norm_out.eps)
+        return self.conv_out(x)
+
+    forward = decode

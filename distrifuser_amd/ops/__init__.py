@@ -1,0 +1,28 @@
+"""Device ops: hand-written gfx950 HIP kernels with eager CPU references.
+
+Dispatch rule: on a ROCm GPU the HIP extension is REQUIRED — a missing
+extension raises instead of silently falling back to eager PyTorch (so GPU
+runs always exercise the native kernels). On CPU the eager reference path
+runs, which is what the non-GPU test suite checks numerics against.
+Set DFA_FORCE_EAGER=1 to force the eager path on GPU (A/B debugging only).
+"""
+
+from .dispatch import (
+    flash_attention,
+    geglu,
+    group_norm_apply,
+    group_norm_silu,
+    group_norm_stats,
+    hip_ext,
+    hip_ext_available,
+)
+
+__all__ = [
+    "flash_attention",
+    "geglu",
+    "group_norm_apply",
+    "group_norm_silu",
+    "group_norm_stats",
+    "hip_ext",
+    "hip_ext_available",
+]

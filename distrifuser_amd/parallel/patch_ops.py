@@ -1,0 +1,412 @@
+"""Displaced-patch-parallel layers (native, not monkey-patched).
+
+These implement the reference's patch-parallel layer semantics
+(/root/reference/distrifuser/modules/pp/{conv2d,attn,groupnorm}.py) as
+first-class modules of our own model stack:
+
+* ``PatchConv2d`` — halo conv: exchanges the ``padding`` boundary rows with
+  the two patch neighbours; stale halos in steady state, synchronous during
+  warmup / full_sync.
+* ``PatchSelfAttention`` — local-query / full-stale-KV attention. The fresh
+  local KV slice is staged into this rank's slot of the flat comm buffer, so
+  the full KV is consumed as a strided VIEW of the buffer (no torch.cat on
+  the hot path — the gfx950 flash-attention kernel walks the per-peer
+  chunks directly).
+* ``PatchGroupNorm`` — six staleness modes incl. the corrected-async
+  estimator full = mean(stale) + (fresh_local - stale_own); SiLU optionally
+  fused into the normalization epilogue.
+* ``CachedCrossAttention`` — text KV computed once per generation (step 0)
+  and cached; no communication.
+
+Every module shares one ``ParallelState`` (counter / comm manager / config).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+from torch import nn
+
+from .. import ops
+from .state import ParallelState
+
+
+def _is_patch_parallel(state: ParallelState) -> bool:
+    return state.config.parallelism == "patch" and state.config.n_device_per_batch > 1
+
+
+class PatchConv2d(nn.Module):
+    """Conv2d whose input is a row-band patch of the full feature map.
+
+    ``is_first_layer`` (conv_in): input is the FULL latent; each rank slices
+    the row band that produces its own output band (plus receptive-field
+    overlap) — semantics of reference pp/conv2d.py:20-41.
+    Interior convs: input is the local band; the ``padding`` top/bottom halo
+    rows come from the patch neighbours' (possibly stale) activations
+    (reference pp/conv2d.py:59-112).
+    """
+
+    def __init__(
+        self,
+        in_channels: int,
+        out_channels: int,
+        kernel_size: int = 3,
+        stride: int = 1,
+        padding: int = 1,
+        bias: bool = True,
+        *,
+        state: ParallelState,
+        is_first_layer: bool = False,
+    ):
+        super().__init__()
+        self.conv = nn.Conv2d(
+            in_channels, out_channels, kernel_size, stride=stride, padding=padding, bias=bias
+        )
+        self.state = state
+        self.is_first_layer = is_first_layer
+        self._idx: int | None = None
+        self._buffer_list: list[torch.Tensor] | None = None
+
+    def reset(self) -> None:
+        self._idx = None
+        self._buffer_list = None
+
+    # -- paths -------------------------------------------------------------
+
+    def _sliced_forward(self, x: torch.Tensor) -> torch.Tensor:
+        """conv_in: compute only this rank's output band from the full input."""
+        cfg = self.state.config
+        b, c, h, w = x.shape
+        n = cfg.n_device_per_batch
+        stride = self.conv.stride[0]
+        padding = self.conv.padding[0]
+        assert h % (stride * n) == 0, f"latent H={h} must divide stride*n={stride * n}"
+
+        output_h = h // stride // n
+        idx = cfg.split_idx()
+        h_begin = output_h * idx * stride - padding
+        h_end = output_h * (idx + 1) * stride + padding
+        pad = [padding, padding, 0, 0]  # W-left, W-right, H-top, H-bottom
+        if h_begin < 0:
+            h_begin = 0
+            pad[2] = padding
+        if h_end > h:
+            h_end = h
+            pad[3] = padding
+        sliced = F.pad(x[:, :, h_begin:h_end, :], pad)
+        return F.conv2d(sliced, self.conv.weight, self.conv.bias, stride=stride)
+
+    def _halo_forward(self, x: torch.Tensor) -> torch.Tensor:
+        cfg = self.state.config
+        comm = self.state.comm_manager
+        halo = self.conv.padding[0]
+
+        if self._buffer_list is None:
+            if comm is None or comm.buffer is None:
+                if comm is not None and self.state.recording and self._idx is None:
+                    self._idx = comm.register_tensor(
+                        (2, x.shape[0], x.shape[1], halo, x.shape[3]),
+                        x.dtype,
+                        layer_type="conv2d",
+                    )
+                # Registration pass: shapes only; halo numerics don't matter yet.
+                return self.conv(x)
+            if self._idx is None:
+                return self.conv(x)
+            self._buffer_list = comm.get_buffer_list(self._idx)
+
+        comm.wait(self._idx)
+
+        # Fresh boundary: [top rows, bottom rows] stacked -> [2, B, C, halo, W]
+        boundary = torch.stack([x[:, :, :halo, :], x[:, :, -halo:, :]], dim=0)
+
+        if self.state.use_sync_comm:
+            dist.all_gather(self._buffer_list, boundary, group=cfg.batch_group)
+        split = cfg.split_idx()
+        n = cfg.n_device_per_batch
+        parts = []
+        if split == 0:
+            parts.append(x.new_zeros(x.shape[0], x.shape[1], halo, x.shape[3]))
+        else:
+            parts.append(self._buffer_list[split - 1][1])  # neighbour-above's bottom rows
+        parts.append(x)
+        if split == n - 1:
+            parts.append(x.new_zeros(x.shape[0], x.shape[1], halo, x.shape[3]))
+        else:
+            parts.append(self._buffer_list[split + 1][0])  # neighbour-below's top rows
+        padded = torch.cat(parts, dim=2)
+        out = F.conv2d(
+            padded,
+            self.conv.weight,
+            self.conv.bias,
+            stride=self.conv.stride[0],
+            padding=(0, self.conv.padding[1]),
+        )
+        if not self.state.use_sync_comm and cfg.mode != "no_sync":
+            comm.enqueue(self._idx, boundary)
+        return out
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if not _is_patch_parallel(self.state):
+            return self.conv(x)
+        if self.is_first_layer:
+            return self._sliced_forward(x)
+        if self.conv.padding[0] == 0:
+            return self.conv(x)  # 1x1: purely local
+        return self._halo_forward(x)
+
+
+class PatchGroupNorm(nn.Module):
+    """GroupNorm over the full spatial extent from per-patch statistics."""
+
+    def __init__(
+        self,
+        num_groups: int,
+        num_channels: int,
+        eps: float = 1e-5,
+        affine: bool = True,
+        *,
+        state: ParallelState,
+        fuse_silu: bool = False,
+    ):
+        super().__init__()
+        self.num_groups = num_groups
+        self.num_channels = num_channels
+        self.eps = eps
+        self.affine = affine
+        self.fuse_silu = fuse_silu
+        if affine:
+            self.weight = nn.Parameter(torch.ones(num_channels))
+            self.bias = nn.Parameter(torch.zeros(num_channels))
+        else:
+            self.register_parameter("weight", None)
+            self.register_parameter("bias", None)
+        self.state = state
+        self._idx: int | None = None
+        self._buffer_list: list[torch.Tensor] | None = None
+
+    def reset(self) -> None:
+        self._idx = None
+        self._buffer_list = None
+
+    def _local(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.group_norm_silu(
+            x, self.num_groups, self.weight, self.bias, self.eps, silu=self.fuse_silu
+        )
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        state = self.state
+        cfg = state.config
+        if not _is_patch_parallel(state):
+            return self._local(x)
+        mode = cfg.mode
+
+        if mode in ("separate_gn", "no_sync"):
+            return self._local(x)
+
+        if mode in ("sync_gn", "full_sync"):
+            # Every-step all-reduce of the stacked [E[x], E[x^2]].
+            stats = ops.group_norm_stats(x, self.num_groups)
+            dist.all_reduce(stats, op=dist.ReduceOp.SUM, group=cfg.batch_group)
+            stats = stats / cfg.n_device_per_batch
+            return ops.group_norm_apply(
+                x, stats[0], stats[1], self.weight, self.bias, self.eps, silu=self.fuse_silu
+            )
+
+        assert mode in ("stale_gn", "corrected_async_gn")
+        comm = state.comm_manager
+        n, c, h, w = x.shape
+
+        if self._buffer_list is None:
+            if comm is None or comm.buffer is None:
+                if comm is not None and state.recording and self._idx is None:
+                    self._idx = comm.register_tensor(
+                        (2, n, self.num_groups, 1, 1, 1), x.dtype, layer_type="gn"
+                    )
+                return self._local(x)
+            if self._idx is None:
+                return self._local(x)
+            self._buffer_list = comm.get_buffer_list(self._idx)
+
+        comm.wait(self._idx)
+        fresh = ops.group_norm_stats(x, self.num_groups)  # [2, N, G, 1, 1, 1]
+
+        if state.in_warmup:
+            dist.all_gather(self._buffer_list, fresh, group=cfg.batch_group)
+            full = torch.stack(self._buffer_list).float().mean(dim=0)
+        elif mode == "corrected_async_gn":
+            # Unbiased corrected estimator: mean(stale) + (fresh - stale_own).
+            stale_own = self._buffer_list[cfg.split_idx()]
+            full = torch.stack(self._buffer_list).float().mean(dim=0) + (
+                fresh.float() - stale_own.float()
+            )
+            comm.enqueue(self._idx, fresh)
+        else:  # stale_gn: substitute own fresh slot
+            stack = torch.stack(
+                [
+                    fresh if p == cfg.split_idx() else buf
+                    for p, buf in enumerate(self._buffer_list)
+                ]
+            )
+            full = stack.float().mean(dim=0)
+            comm.enqueue(self._idx, fresh)
+
+        mean, meansq = full[0], full[1]
+        if mode == "corrected_async_gn" and not state.in_warmup:
+            # The corrected estimator can produce E[x^2] - E[x]^2 < 0; fall
+            # back to the local-slice variance there (reference
+            # pp/groupnorm.py:60-63). Implemented by clamping meansq up to
+            # the local value where the corrected variance goes negative.
+            var = meansq - mean * mean
+            f_mean, f_meansq = fresh[0].float(), fresh[1].float()
+            local_var = f_meansq - f_mean * f_mean
+            neg = var < 0
+            meansq = torch.where(neg, mean * mean + local_var, meansq)
+        return ops.group_norm_apply(
+            x, mean, meansq, self.weight, self.bias, self.eps, silu=self.fuse_silu
+        )
+
+
+class PatchSelfAttention(nn.Module):
+    """Local-query / full-(stale-)KV self-attention.
+
+    K and V projections are fused into one ``to_kv`` GEMM (one collective +
+    one GEMM instead of two — reference pp/attn.py:23-39). In steady state
+    the full-sequence KV is the flat comm buffer itself: the fresh local
+    slice is copied into this rank's slot (which the enqueue needs anyway),
+    and the kernel consumes the buffer as a [n_peers, B, L_local, 2C]
+    strided view.
+    """
+
+    def __init__(
+        self,
+        query_dim: int,
+        heads: int,
+        dim_head: int,
+        out_bias: bool = True,
+        *,
+        state: ParallelState,
+    ):
+        super().__init__()
+        inner = heads * dim_head
+        self.heads = heads
+        self.dim_head = dim_head
+        self.to_q = nn.Linear(query_dim, inner, bias=False)
+        self.to_kv = nn.Linear(query_dim, 2 * inner, bias=False)
+        self.to_out = nn.Linear(inner, query_dim, bias=out_bias)
+        self.state = state
+        self._idx: int | None = None
+        self._buffer_view: torch.Tensor | None = None  # [n, B, L, 2*inner]
+        self._buffer_list: list[torch.Tensor] | None = None
+
+    def reset(self) -> None:
+        self._idx = None
+        self._buffer_view = None
+        self._buffer_list = None
+
+    def _attention(self, q: torch.Tensor, kv: torch.Tensor) -> torch.Tensor:
+        """q: [B, Lq, inner]; kv: [B, Lkv, 2*inner] (may be strided)."""
+        b, lq, inner = q.shape
+        lkv = kv.shape[1]
+        k, v = kv.split(inner, dim=-1)
+        q = q.view(b, lq, self.heads, self.dim_head).transpose(1, 2)
+        k = k.view(b, lkv, self.heads, self.dim_head).transpose(1, 2)
+        v = v.view(b, lkv, self.heads, self.dim_head).transpose(1, 2)
+        out = ops.flash_attention(q, k, v)
+        return out.transpose(1, 2).reshape(b, lq, inner)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        state = self.state
+        cfg = state.config
+        b, l, _ = x.shape
+        q = self.to_q(x)
+        kv = self.to_kv(x)
+
+        if not _is_patch_parallel(state):
+            out = self._attention(q, kv)
+            return self.to_out(out)
+
+        comm = state.comm_manager
+        if self._buffer_view is None:
+            if comm is None or comm.buffer is None or self._idx is None:
+                if (
+                    comm is not None
+                    and comm.buffer is None
+                    and state.recording
+                    and self._idx is None
+                ):
+                    self._idx = comm.register_tensor(
+                        (b, l, self.to_kv.out_features), x.dtype, layer_type="attn"
+                    )
+                # Registration pass: fake the full-sequence shape.
+                full_kv = kv.repeat(1, cfg.n_device_per_batch, 1)
+                return self.to_out(self._attention(q, full_kv))
+            self._buffer_list = comm.get_buffer_list(self._idx)
+            n = cfg.n_device_per_batch
+            s, e = comm.starts[self._idx], comm.ends[self._idx]
+            self._buffer_view = comm.buffer[:, s:e].view(n, b, l, self.to_kv.out_features)
+
+        comm.wait(self._idx)
+        own = cfg.split_idx()
+        if state.use_sync_comm:
+            dist.all_gather(self._buffer_list, kv, group=cfg.batch_group)
+        else:
+            # Stage the fresh slice into our slot: the buffer then holds
+            # [stale..., fresh_own, ...stale] — exactly the displaced KV.
+            self._buffer_list[own].copy_(kv)
+            if cfg.mode != "no_sync":
+                comm.enqueue(self._idx)  # fresh slice already staged in-slot
+
+        # [n, B, L, 2C] -> [B, n*L, 2C]; peers are ordered by patch index so
+        # the concatenated sequence is the full image in row order.
+        full_kv = self._buffer_view.permute(1, 0, 2, 3).reshape(
+            b, cfg.n_device_per_batch * l, self.to_kv.out_features
+        )
+        return self.to_out(self._attention(q, full_kv))
+
+
+class CachedCrossAttention(nn.Module):
+    """Text-conditioned cross-attention with a per-generation KV cache.
+
+    encoder_hidden_states are constant across denoise steps, so KV is
+    computed once at counter==0 and cached (reference pp/attn.py:42-104);
+    queries are per-patch, so no communication at all.
+    """
+
+    def __init__(
+        self,
+        query_dim: int,
+        cross_dim: int,
+        heads: int,
+        dim_head: int,
+        out_bias: bool = True,
+        *,
+        state: ParallelState,
+    ):
+        super().__init__()
+        inner = heads * dim_head
+        self.heads = heads
+        self.dim_head = dim_head
+        self.to_q = nn.Linear(query_dim, inner, bias=False)
+        self.to_kv = nn.Linear(cross_dim, 2 * inner, bias=False)
+        self.to_out = nn.Linear(inner, query_dim, bias=out_bias)
+        self.state = state
+        self.kv_cache: torch.Tensor | None = None
+
+    def reset(self) -> None:
+        self.kv_cache = None
+
+    def forward(self, x: torch.Tensor, encoder_hidden_states: torch.Tensor) -> torch.Tensor:
+        b, lq, _ = x.shape
+        if self.state.counter == 0 or self.kv_cache is None:
+            self.kv_cache = self.to_kv(encoder_hidden_states)
+        kv = self.kv_cache
+        inner = self.heads * self.dim_head
+        k, v = kv.split(inner, dim=-1)
+        lkv = kv.shape[1]
+        q = self.to_q(x).view(b, lq, self.heads, self.dim_head).transpose(1, 2)
+        k = k.view(b, lkv, self.heads, self.dim_head).transpose(1, 2)
+        v = v.view(b, lkv, self.heads, self.dim_head).transpose(1, 2)
+        out = ops.flash_attention(q, k, v).transpose(1, 2).reshape(b, lq, inner)
+        return self.to_out(out)

@@ -1,0 +1,395 @@
+"""DistriSDXLPipeline / DistriSDPipeline — the user-facing API.
+
+Parity with the reference surface (reference pipelines.py): construction via
+``from_pretrained(distri_config, **kw)``, generation via ``pipeline(prompt=
+..., generator=...)``, ``prepare()`` runs the comm-buffer registration pass,
+the buffer-filling pre-run, and hipGraph capture. Unlike the reference
+(which wrapped diffusers' StableDiffusion*Pipeline), the scheduler loop,
+prompt encoding, latent handling and VAE decode here are all owned natively.
+
+No network in this environment: ``pretrained_model_name_or_path=None`` (the
+default) builds random-init weights of the exact architecture; a local
+diffusers-layout directory with safetensors files is loaded when given.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .models.clip import CLIP_VIT_L, OPEN_CLIP_BIG_G, TINY_CLIP, CLIPTextEncoder
+from .models.distri_unet import DistriUNet
+from .models.tokenizer import SimpleTokenizer
+from .models.unet import SD15_UNET, SDXL_UNET, TINY_UNET, UNetConfig
+from .models.vae import SD_VAE, SDXL_VAE, TINY_VAE, VAEDecoder, VAEDecoderConfig
+from .models import weights as weight_io
+from .schedulers import get_scheduler
+from .utils.comm import PatchParallelismCommManager
+from .utils.config import DistriConfig
+
+
+def _maybe_tqdm(iterable, enabled: bool):
+    if not enabled:
+        return iterable
+    try:
+        from tqdm import tqdm
+
+        return tqdm(iterable)
+    except ImportError:  # pragma: no cover
+        return iterable
+
+
+class _DistriPipelineBase:
+    is_sdxl = False
+
+    def __init__(
+        self,
+        distri_config: DistriConfig,
+        unet: DistriUNet,
+        vae: VAEDecoder,
+        scheduler,
+        tokenizer: SimpleTokenizer,
+    ):
+        self.distri_config = distri_config
+        self.unet = unet
+        self.vae = vae
+        self.scheduler = scheduler
+        self.tokenizer = tokenizer
+        self.comm_manager: PatchParallelismCommManager | None = None
+        self._progress = distri_config.rank == 0 and distri_config.verbose
+        self.prepare()
+
+    # -- reference-API conveniences -----------------------------------------
+
+    def set_progress_bar_config(self, disable: bool = False, **kwargs):
+        self._progress = not disable and self.distri_config.rank == 0
+
+    @property
+    def device(self):
+        return self.distri_config.device
+
+    # -- core helpers --------------------------------------------------------
+
+    def _unet_dtype(self) -> torch.dtype:
+        return next(self.unet.parameters()).dtype
+
+    def _prepare_latents(self, batch_size: int, generator) -> torch.Tensor:
+        cfg = self.distri_config
+        shape = (batch_size, self.unet.config.in_channels, cfg.height // 8, cfg.width // 8)
+        if generator is None:
+            # Cross-rank determinism: every rank MUST sample identical initial
+            # latents (they denoise the same image). Default to a fixed seed.
+            generator = torch.Generator().manual_seed(0)
+        device = generator.device if hasattr(generator, "device") else torch.device("cpu")
+        latents = torch.randn(shape, generator=generator, dtype=torch.float32, device=device)
+        latents = latents.to(device=cfg.device, dtype=self._unet_dtype())
+        return latents * self.scheduler.init_noise_sigma
+
+    def _denoise(self, latents, prompt_embeds, added_cond_kwargs, num_inference_steps, guidance_scale):
+        cfg = self.distri_config
+        do_cfg = cfg.do_classifier_free_guidance
+        self.scheduler.set_timesteps(num_inference_steps, device=None)
+        self.unet.set_counter(0)
+        for t in _maybe_tqdm(self.scheduler.timesteps, self._progress):
+            latent_in = torch.cat([latents] * 2) if do_cfg else latents
+            latent_in = self.scheduler.scale_model_input(latent_in, t)
+            noise = self.unet(
+                latent_in,
+                t.to(cfg.device) if torch.is_tensor(t) else t,
+                prompt_embeds,
+                added_cond_kwargs,
+            )
+            if do_cfg:
+                n_uncond, n_cond = noise.chunk(2)
+                noise = n_uncond + guidance_scale * (n_cond - n_uncond)
+            latents = self.scheduler.step(noise, t, latents)
+        return latents
+
+    def _decode(self, latents: torch.Tensor, output_type: str):
+        if output_type == "latent":
+            return latents
+        images = self.vae.decode(latents.to(self._unet_dtype()))
+        if output_type == "pt":
+            return images
+        arr = ((images.float() / 2 + 0.5).clamp(0, 1) * 255).round().to(torch.uint8)
+        arr = arr.permute(0, 2, 3, 1).cpu().numpy()
+        if output_type == "np":
+            return arr
+        try:  # "pil"
+            from PIL import Image
+
+            return [Image.fromarray(a) for a in arr]
+        except ImportError:
+            return arr  # no pillow in this environment; numpy HWC uint8
+
+    # -- prepare: registration pass, comm buffer, pre-run, hipGraph capture --
+
+    @torch.no_grad()
+    def prepare(self):
+        cfg = self.distri_config
+        unet = self.unet
+
+        needs_comm = cfg.parallelism == "patch" and cfg.n_device_per_batch > 1
+        wants_graphs = cfg.use_cuda_graph and cfg.device.type == "cuda"
+        if not needs_comm and not wants_graphs:
+            return  # nothing to warm up
+        batch_size = 2 if cfg.do_classifier_free_guidance else 1
+
+        static = self._build_static_inputs(batch_size)
+
+        if needs_comm:
+            self.comm_manager = PatchParallelismCommManager(cfg)
+            unet.set_comm_manager(self.comm_manager)
+            # registration pass: sizes every stale-activation slot
+            unet.set_counter(0)
+            unet(**static, record=True)
+            self.comm_manager.create_buffer()
+
+        # pre-run: modules grab their buffer views and fill them
+        unet.set_counter(0)
+        unet(**static, record=True)
+        if self.comm_manager is not None:
+            self.comm_manager.clear()
+
+        if cfg.use_cuda_graph and cfg.device.type == "cuda":
+            self._capture_graphs(static)
+
+    def _graph_counters(self) -> list[int]:
+        cfg = self.distri_config
+        if cfg.world_size == 1 or cfg.n_device_per_batch == 1:
+            return [0]
+        if cfg.parallelism == "patch":
+            return [0, cfg.warmup_steps + 1, cfg.warmup_steps + 2]
+        if cfg.parallelism == "naive_patch" and cfg.split_scheme == "alternate":
+            return [0, 1]
+        return [0]
+
+    def _capture_graphs(self, static):
+        unet = self.unet
+        graphs, outputs = [], []
+        try:
+            torch.cuda.synchronize()
+            for counter in self._graph_counters():
+                g = torch.cuda.CUDAGraph()
+                unet.set_counter(counter)
+                with torch.cuda.graph(g):
+                    out = unet(**static, record=True)
+                graphs.append(g)
+                outputs.append(out)
+            unet.setup_cuda_graph(outputs, graphs)
+        except Exception as exc:  # pragma: no cover - depends on RCCL graph support
+            if self.distri_config.rank == 0:
+                print(f"[distrifuser_amd] hipGraph capture failed ({exc}); running eager")
+            unet.setup_cuda_graph(None, None)
+            unet.cuda_graphs = None
+            if self.comm_manager is not None:
+                self.comm_manager.clear()
+
+    def _build_static_inputs(self, batch_size: int) -> dict:
+        raise NotImplementedError
+
+
+class DistriSDXLPipeline(_DistriPipelineBase):
+    is_sdxl = True
+
+    def __init__(self, distri_config, unet, vae, text_encoder, text_encoder_2, scheduler, tokenizer):
+        self.text_encoder = text_encoder
+        self.text_encoder_2 = text_encoder_2
+        super().__init__(distri_config, unet, vae, scheduler, tokenizer)
+
+    @staticmethod
+    def from_pretrained(distri_config: DistriConfig, **kwargs):
+        pretrained = kwargs.pop("pretrained_model_name_or_path", None)
+        torch_dtype = kwargs.pop("torch_dtype", torch.bfloat16)
+        scheduler = kwargs.pop("scheduler", "ddim")
+        preset = kwargs.pop("preset", "sdxl")
+        device = distri_config.device
+
+        if preset == "tiny":
+            unet_cfg, vae_cfg = TINY_UNET, TINY_VAE
+            clip1_cfg, clip2_cfg = TINY_CLIP, TINY_CLIP
+            tokenizer = SimpleTokenizer(vocab_size=TINY_CLIP.vocab_size)
+        else:
+            unet_cfg, vae_cfg = SDXL_UNET, SDXL_VAE
+            clip1_cfg, clip2_cfg = CLIP_VIT_L, OPEN_CLIP_BIG_G
+            tokenizer = SimpleTokenizer()
+        unet_cfg = kwargs.pop("unet_config", unet_cfg)
+
+        unet = DistriUNet(unet_cfg, distri_config)
+        vae = VAEDecoder(vae_cfg)
+        te1 = CLIPTextEncoder(clip1_cfg)
+        te2 = CLIPTextEncoder(clip2_cfg)
+
+        if pretrained is not None:
+            for model, comp in ((unet.unet, "unet"), (vae, "vae"),
+                                (te1, "text_encoder"), (te2, "text_encoder_2")):
+                path = weight_io.find_component_weights(pretrained, comp)
+                if path is not None:
+                    weight_io.load_into(model, weight_io.load_safetensors(path))
+
+        unet = unet.to(device=device, dtype=torch_dtype).eval()
+        vae = vae.to(device=device, dtype=torch_dtype).eval()
+        te1 = te1.to(device=device, dtype=torch_dtype).eval()
+        te2 = te2.to(device=device, dtype=torch_dtype).eval()
+        return DistriSDXLPipeline(distri_config, unet, vae, te1, te2,
+                                  get_scheduler(scheduler), tokenizer)
+
+    @torch.no_grad()
+    def encode_prompt(self, prompt, negative_prompt=None, do_classifier_free_guidance=True):
+        """Returns (prompt_embeds [B,77,2048], pooled [B,1280]) per branch,
+        concatenated [uncond; cond] when CFG is on."""
+        device = self.distri_config.device
+        ids1 = self.tokenizer(prompt, device=device)
+        emb1, _ = self.text_encoder(ids1, hidden_state_index=-2)
+        emb2, pooled = self.text_encoder_2(ids1, hidden_state_index=-2)
+        embeds = torch.cat([emb1, emb2], dim=-1)
+        if not do_classifier_free_guidance:
+            return embeds, pooled
+        neg = negative_prompt if negative_prompt is not None else ""
+        nids = self.tokenizer(neg, device=device)
+        nemb1, _ = self.text_encoder(nids, hidden_state_index=-2)
+        nemb2, npooled = self.text_encoder_2(nids, hidden_state_index=-2)
+        nembeds = torch.cat([nemb1, nemb2], dim=-1)
+        return torch.cat([nembeds, embeds]), torch.cat([npooled, pooled])
+
+    def _added_cond(self, batch_size: int, pooled: torch.Tensor) -> dict:
+        cfg = self.distri_config
+        time_ids = torch.tensor(
+            [[cfg.height, cfg.width, 0, 0, cfg.height, cfg.width]],
+            device=cfg.device, dtype=pooled.dtype,
+        ).repeat(batch_size, 1)
+        return {"text_embeds": pooled, "time_ids": time_ids}
+
+    def _build_static_inputs(self, batch_size: int) -> dict:
+        embeds, pooled = self.encode_prompt("", do_classifier_free_guidance=False)
+        embeds = embeds.repeat(batch_size, 1, 1).to(self._unet_dtype())
+        pooled = pooled.repeat(batch_size, 1).to(self._unet_dtype())
+        latents = self._prepare_latents(batch_size, None)
+        return {
+            "sample": latents,
+            "timestep": torch.zeros((), dtype=torch.float32, device=self.distri_config.device),
+            "encoder_hidden_states": embeds,
+            "added_cond_kwargs": self._added_cond(batch_size, pooled),
+        }
+
+    @torch.no_grad()
+    def __call__(
+        self,
+        prompt: str = "",
+        negative_prompt: str | None = None,
+        num_inference_steps: int = 50,
+        guidance_scale: float = 5.0,
+        generator: torch.Generator | None = None,
+        output_type: str = "pil",
+        **kwargs,
+    ):
+        assert "height" not in kwargs and "width" not in kwargs, (
+            "height/width are fixed by DistriConfig (the comm buffers and "
+            "graphs are sized for them)"
+        )
+        cfg = self.distri_config
+        if not cfg.do_classifier_free_guidance:
+            assert guidance_scale == 1 or guidance_scale is None
+            guidance_scale = 1
+        do_cfg = cfg.do_classifier_free_guidance
+
+        embeds, pooled = self.encode_prompt(prompt, negative_prompt, do_cfg)
+        embeds = embeds.to(self._unet_dtype())
+        pooled = pooled.to(self._unet_dtype())
+        added = self._added_cond(embeds.shape[0], pooled)
+        latents = self._prepare_latents(1, generator)
+        latents = self._denoise(latents, embeds, added, num_inference_steps, guidance_scale)
+        return self._decode(latents, output_type)
+
+
+class DistriSDPipeline(_DistriPipelineBase):
+    """SD 1.x/2.x pipeline (single text encoder, no added conditions)."""
+
+    def __init__(self, distri_config, unet, vae, text_encoder, scheduler, tokenizer):
+        self.text_encoder = text_encoder
+        super().__init__(distri_config, unet, vae, scheduler, tokenizer)
+
+    @staticmethod
+    def from_pretrained(distri_config: DistriConfig, **kwargs):
+        pretrained = kwargs.pop("pretrained_model_name_or_path", None)
+        torch_dtype = kwargs.pop("torch_dtype", torch.bfloat16)
+        scheduler = kwargs.pop("scheduler", "ddim")
+        preset = kwargs.pop("preset", "sd15")
+        device = distri_config.device
+
+        if preset == "tiny":
+            unet_cfg = UNetConfig(
+                block_out_channels=(32, 64),
+                down_block_types=("CrossAttnDownBlock2D", "DownBlock2D"),
+                layers_per_block=1,
+                transformer_layers_per_block=(1, 1),
+                num_attention_heads=(2, 4),
+                cross_attention_dim=16,
+                norm_num_groups=8,
+                use_linear_projection=False,
+                addition_embed_type=None,
+                sample_size=8,
+            )
+            vae_cfg, clip_cfg = TINY_VAE, TINY_CLIP
+            tokenizer = SimpleTokenizer(vocab_size=TINY_CLIP.vocab_size)
+        else:
+            unet_cfg, vae_cfg, clip_cfg = SD15_UNET, SD_VAE, CLIP_VIT_L
+            tokenizer = SimpleTokenizer()
+        unet_cfg = kwargs.pop("unet_config", unet_cfg)
+
+        unet = DistriUNet(unet_cfg, distri_config)
+        vae = VAEDecoder(vae_cfg)
+        te = CLIPTextEncoder(clip_cfg)
+        if pretrained is not None:
+            for model, comp in ((unet.unet, "unet"), (vae, "vae"), (te, "text_encoder")):
+                path = weight_io.find_component_weights(pretrained, comp)
+                if path is not None:
+                    weight_io.load_into(model, weight_io.load_safetensors(path))
+
+        unet = unet.to(device=device, dtype=torch_dtype).eval()
+        vae = vae.to(device=device, dtype=torch_dtype).eval()
+        te = te.to(device=device, dtype=torch_dtype).eval()
+        return DistriSDPipeline(distri_config, unet, vae, te, get_scheduler(scheduler), tokenizer)
+
+    @torch.no_grad()
+    def encode_prompt(self, prompt, negative_prompt=None, do_classifier_free_guidance=True):
+        device = self.distri_config.device
+        ids = self.tokenizer(prompt, device=device)
+        emb, _ = self.text_encoder(ids, hidden_state_index=-1)
+        if not do_classifier_free_guidance:
+            return emb
+        neg = negative_prompt if negative_prompt is not None else ""
+        nids = self.tokenizer(neg, device=device)
+        nemb, _ = self.text_encoder(nids, hidden_state_index=-1)
+        return torch.cat([nemb, emb])
+
+    def _build_static_inputs(self, batch_size: int) -> dict:
+        embeds = self.encode_prompt("", do_classifier_free_guidance=False)
+        embeds = embeds.repeat(batch_size, 1, 1).to(self._unet_dtype())
+        latents = self._prepare_latents(batch_size, None)
+        return {
+            "sample": latents,
+            "timestep": torch.zeros((), dtype=torch.float32, device=self.distri_config.device),
+            "encoder_hidden_states": embeds,
+            "added_cond_kwargs": None,
+        }
+
+    @torch.no_grad()
+    def __call__(
+        self,
+        prompt: str = "",
+        negative_prompt: str | None = None,
+        num_inference_steps: int = 50,
+        guidance_scale: float = 7.5,
+        generator: torch.Generator | None = None,
+        output_type: str = "pil",
+        **kwargs,
+    ):
+        assert "height" not in kwargs and "width" not in kwargs
+        cfg = self.distri_config
+        if not cfg.do_classifier_free_guidance:
+            guidance_scale = 1
+        do_cfg = cfg.do_classifier_free_guidance
+        embeds = self.encode_prompt(prompt, negative_prompt, do_cfg).to(self._unet_dtype())
+        latents = self._prepare_latents(1, generator)
+        latents = self._denoise(latents, embeds, None, num_inference_steps, guidance_scale)
+        return self._decode(latents, output_type)

@@ -1,0 +1,174 @@
+"""The built-in correctness oracle: full_sync patch parallelism
+must match the single-process U-Net output exactly, and with a static input
+the displaced (stale) modes equal full_sync after warmup (stale == fresh when
+the input never changes)."""
+
+import torch
+
+from distrifuser_amd import DistriConfig, PatchParallelismCommManager
+from distrifuser_amd.models import DistriUNet
+from distrifuser_amd.models.unet import TINY_UNET
+
+from conftest import run_distributed
+
+H = W = 16  # latent size for the tiny config
+
+
+def _tiny_inputs(batch):
+    g = torch.Generator().manual_seed(42)
+    sample = torch.randn(batch, 4, H, W, generator=g)
+    ehs = torch.randn(batch, 7, TINY_UNET.cross_attention_dim, generator=g)
+    pooled = torch.randn(batch, 16, generator=g)
+    time_ids = torch.tensor([[H * 8, W * 8, 0, 0, H * 8, W * 8]], dtype=torch.float32).repeat(
+        batch, 1
+    )
+    added = {"text_embeds": pooled, "time_ids": time_ids}
+    return sample, ehs, added
+
+
+def _build_unet(mode="full_sync", parallelism="patch", do_cfg=False, warmup=2):
+    cfg = DistriConfig(
+        height=H * 8,
+        width=W * 8,
+        do_classifier_free_guidance=do_cfg,
+        mode=mode,
+        parallelism=parallelism,
+        warmup_steps=warmup,
+        use_cuda_graph=False,
+        device="cpu",
+    )
+    torch.manual_seed(0)
+    unet = DistriUNet(TINY_UNET, cfg)
+    unet.eval()
+    return cfg, unet
+
+
+def _prep(cfg, unet, static):
+    if cfg.parallelism == "patch" and cfg.n_device_per_batch > 1:
+        comm = PatchParallelismCommManager(cfg)
+        unet.set_comm_manager(comm)
+        unet.set_counter(0)
+        unet(*static, record=True)
+        comm.create_buffer()
+        unet.set_counter(0)
+        unet(*static, record=True)
+        comm.clear()
+
+
+@torch.no_grad()
+def _single_reference(batch, steps=1):
+    cfg, unet = _build_unet(do_cfg=batch == 2)
+    sample, ehs, added = _tiny_inputs(batch)
+    unet.set_counter(0)
+    outs = [unet(sample, 3.0, ehs, added) for _ in range(steps)]
+    return outs
+
+
+@torch.no_grad()
+def _patch_worker(rank, world_size, mode, do_cfg, steps):
+    cfg, unet = _build_unet(mode=mode, do_cfg=do_cfg)
+    batch = 2 if do_cfg else 1
+    sample, ehs, added = _tiny_inputs(batch)
+    static = (sample, 3.0, ehs, added)
+    _prep(cfg, unet, static)
+    unet.set_counter(0)
+    outs = [unet(sample, 3.0, ehs, added) for _ in range(steps)]
+    return [o.clone() for o in outs]
+
+
+def test_unet_shapes_single():
+    (out,) = _single_reference(1)
+    assert out.shape == (1, 4, H, W)
+    assert torch.isfinite(out).all()
+
+
+def test_patch_full_sync_matches_single_ws2():
+    ref = _single_reference(1)[0]
+    out = run_distributed(2, _patch_worker, ("full_sync", False, 1))
+    for r in (0, 1):
+        assert out[r][0].shape == ref.shape
+        assert torch.allclose(out[r][0], ref, atol=2e-4), (
+            f"max err {(out[r][0] - ref).abs().max()}"
+        )
+
+
+def test_patch_cfg_split_matches_single_ws2():
+    """ws=2 with CFG split: pure batch parallelism (patch degree 1)."""
+    ref = _single_reference(2)[0]
+    out = run_distributed(2, _patch_worker, ("full_sync", True, 1))
+    for r in (0, 1):
+        assert torch.allclose(out[r][0], ref, atol=2e-4)
+
+
+def test_patch_full_sync_matches_single_ws4_cfg():
+    """ws=4: CFG split x 2 patches."""
+    ref = _single_reference(2)[0]
+    out = run_distributed(4, _patch_worker, ("full_sync", True, 1))
+    for r in range(4):
+        assert torch.allclose(out[r][0], ref, atol=2e-4), (
+            f"rank {r} max err {(out[r][0] - ref).abs().max()}"
+        )
+
+
+def test_corrected_async_equals_full_sync_on_static_input():
+    """With an unchanging input, stale activations equal fresh ones, so the
+    displaced mode must converge to full_sync exactly once buffers filled."""
+    steps = 5  # warmup=2, so steps 3.. are steady-state async
+    ref = _single_reference(1, steps=steps)
+    for mode in ("corrected_async_gn", "stale_gn"):
+        out = run_distributed(2, _patch_worker, (mode, False, steps))
+        for r in (0, 1):
+            final = out[r][-1]
+            assert torch.allclose(final, ref[-1], atol=5e-4), (
+                f"{mode}: max err {(final - ref[-1]).abs().max()}"
+            )
+
+
+@torch.no_grad()
+def _naive_worker(rank, world_size, scheme):
+    cfg = DistriConfig(
+        height=H * 8, width=W * 8, do_classifier_free_guidance=False,
+        parallelism="naive_patch", split_scheme=scheme, use_cuda_graph=False, device="cpu",
+    )
+    torch.manual_seed(0)
+    unet = DistriUNet(TINY_UNET, cfg).eval()
+    sample, ehs, added = _tiny_inputs(1)
+    unet.set_counter(0)
+    out = unet(sample, 3.0, ehs, added)
+    return out.clone()
+
+
+def test_naive_patch_shapes_ws2():
+    for scheme in ("row", "col"):
+        out = run_distributed(2, _naive_worker, (scheme,))
+        assert out[0].shape == (1, 4, H, W)
+        assert torch.allclose(out[0], out[1])
+
+
+@torch.no_grad()
+def _tensor_worker(rank, world_size, do_cfg):
+    cfg = DistriConfig(
+        height=H * 8, width=W * 8, do_classifier_free_guidance=do_cfg,
+        parallelism="tensor", use_cuda_graph=False, device="cpu",
+    )
+    torch.manual_seed(0)
+    unet = DistriUNet(TINY_UNET, cfg).eval()
+    sample, ehs, added = _tiny_inputs(2 if do_cfg else 1)
+    unet.set_counter(0)
+    out = unet(sample, 3.0, ehs, added)
+    return out.clone()
+
+
+def test_tensor_parallel_runs_ws2():
+    out = run_distributed(2, _tensor_worker, (False,))
+    assert out[0].shape == (1, 4, H, W)
+    assert torch.allclose(out[0], out[1], atol=1e-5)
+    assert torch.isfinite(out[0]).all()
+
+
+def test_tensor_parallel_cfg_split_ws2():
+    """Exercises the CFG-pair exchange the reference left broken
+    (reference distri_sdxl_unet_tp.py:159-162)."""
+    out = run_distributed(2, _tensor_worker, (True,))
+    assert out[0].shape == (2, 4, H, W)
+    assert torch.allclose(out[0], out[1], atol=1e-5)
