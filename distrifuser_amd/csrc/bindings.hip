@@ -1,0 +1,192 @@
+// Python bindings for the distrifuser_amd gfx950 kernels (module _C).
+// Pure HIP + ATen/hip — no CUDA-compat layer; compiled by hipcc via
+// torch.utils.cpp_extension with PYTORCH_ROCM_ARCH=gfx950.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include "kernels.h"
+
+namespace {
+
+int dtype_of(const at::Tensor& t) {
+    switch (t.scalar_type()) {
+        case at::kBFloat16: return DFA_BF16;
+        case at::kHalf: return DFA_F16;
+        case at::kFloat: return DFA_F32;
+        default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+    }
+}
+
+hipStream_t cur_stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+at::Tensor group_norm_stats(const at::Tensor& x_, int64_t num_groups) {
+    TORCH_CHECK(x_.is_cuda() && x_.dim() == 4, "x must be CUDA [N,C,H,W]");
+    auto x = x_.contiguous();
+    const int64_t n = x.size(0), c = x.size(1), hw = x.size(2) * x.size(3);
+    TORCH_CHECK(c % num_groups == 0);
+    const int64_t group_len = (c / num_groups) * hw;
+    const int ngt = (int)(n * num_groups);
+    auto partial = at::zeros({2, (long)ngt}, x.options().dtype(at::kFloat));
+    launch_gn_stats_partial(x.data_ptr(), partial.data_ptr<float>(), group_len, ngt,
+                            dtype_of(x), cur_stream());
+    auto out = at::empty({2, n, num_groups, 1, 1, 1}, x.options());
+    launch_gn_finalize(partial.data_ptr<float>(), out.data_ptr(), 1.0f / (float)group_len, ngt,
+                       dtype_of(x), cur_stream());
+    return out;
+}
+
+at::Tensor group_norm_apply(const at::Tensor& x_, const at::Tensor& mean_,
+                            const at::Tensor& meansq_, const c10::optional<at::Tensor>& w_,
+                            const c10::optional<at::Tensor>& b_, double eps, bool silu) {
+    TORCH_CHECK(x_.is_cuda() && x_.dim() == 4);
+    auto x = x_.contiguous();
+    const int64_t n = x.size(0), c = x.size(1), hw = x.size(2) * x.size(3);
+    auto mean = mean_.to(at::kFloat).contiguous().view({-1});
+    auto meansq = meansq_.to(at::kFloat).contiguous().view({-1});
+    TORCH_CHECK(mean.numel() % n == 0);
+    const int g = (int)(mean.numel() / n);
+    TORCH_CHECK(c % g == 0, "channels ", c, " not divisible by groups ", g);
+    at::Tensor w, b;
+    const void* wp = nullptr;
+    const void* bp = nullptr;
+    if (w_.has_value()) {
+        w = w_->to(x.scalar_type()).contiguous();
+        wp = w.data_ptr();
+    }
+    if (b_.has_value()) {
+        b = b_->to(x.scalar_type()).contiguous();
+        bp = b.data_ptr();
+    }
+    auto y = at::empty_like(x);
+    launch_gn_apply(x.data_ptr(), y.data_ptr(), mean.data_ptr<float>(), meansq.data_ptr<float>(),
+                    wp, bp, (float)eps, hw, (int)c, g, (int)n, silu, dtype_of(x), cur_stream());
+    return y;
+}
+
+at::Tensor group_norm_silu(const at::Tensor& x_, int64_t num_groups,
+                           const c10::optional<at::Tensor>& w_,
+                           const c10::optional<at::Tensor>& b_, double eps, bool silu) {
+    TORCH_CHECK(x_.is_cuda() && x_.dim() == 4);
+    auto x = x_.contiguous();
+    const int64_t n = x.size(0), c = x.size(1), hw = x.size(2) * x.size(3);
+    TORCH_CHECK(c % num_groups == 0);
+    const int64_t group_len = (c / num_groups) * hw;
+    const int ngt = (int)(n * num_groups);
+    auto partial = at::zeros({2, (long)ngt}, x.options().dtype(at::kFloat));
+    launch_gn_stats_partial(x.data_ptr(), partial.data_ptr<float>(), group_len, ngt,
+                            dtype_of(x), cur_stream());
+    // finalize in fp32 (full precision straight into the apply)
+    auto moments = at::empty({2, (long)ngt}, x.options().dtype(at::kFloat));
+    launch_gn_finalize(partial.data_ptr<float>(), moments.data_ptr(), 1.0f / (float)group_len,
+                       ngt, DFA_F32, cur_stream());
+    at::Tensor w, b;
+    const void* wp = nullptr;
+    const void* bp = nullptr;
+    if (w_.has_value()) {
+        w = w_->to(x.scalar_type()).contiguous();
+        wp = w.data_ptr();
+    }
+    if (b_.has_value()) {
+        b = b_->to(x.scalar_type()).contiguous();
+        bp = b.data_ptr();
+    }
+    auto y = at::empty_like(x);
+    launch_gn_apply(x.data_ptr(), y.data_ptr(), moments.data_ptr<float>(),
+                    moments.data_ptr<float>() + ngt, wp, bp, (float)eps, hw, (int)c,
+                    (int)num_groups, (int)n, silu, dtype_of(x), cur_stream());
+    return y;
+}
+
+at::Tensor geglu(const at::Tensor& h_) {
+    TORCH_CHECK(h_.is_cuda());
+    auto h = h_.contiguous();
+    const int64_t inner = h.size(-1) / 2;
+    TORCH_CHECK(h.size(-1) % 2 == 0);
+    auto sizes = h.sizes().vec();
+    sizes.back() = inner;
+    auto out = at::empty(sizes, h.options());
+    launch_geglu(h.data_ptr(), out.data_ptr(), h.numel() / (2 * inner), inner, dtype_of(h),
+                 cur_stream());
+    return out;
+}
+
+at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v) {
+    // q: [B,H,Lq,64]; k/v: [B,H,Lkv,64] or [B,H,NC,LC,64] (stale-KV chunks)
+    TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "flash_attention: bf16 only");
+    TORCH_CHECK(q.dim() == 4 && q.size(-1) == 64, "head_dim must be 64");
+    TORCH_CHECK(q.stride(-1) == 1 && k.stride(-1) == 1 && v.stride(-1) == 1,
+                "innermost dim must be contiguous");
+    const int B = (int)q.size(0), H = (int)q.size(1);
+    const int64_t Lq = q.size(2);
+
+    FlashAttnParams p{};
+    p.q = reinterpret_cast<const uint16_t*>(q.data_ptr());
+    p.B = B;
+    p.H = H;
+    p.Lq = Lq;
+    p.q_sb = q.stride(0);
+    p.q_sh = q.stride(1);
+    p.q_sl = q.stride(2);
+    p.scale = 0.125f;  // 1/sqrt(64)
+
+    auto set_kv = [&](const at::Tensor& t, const uint16_t*& ptr, int64_t& sb, int64_t& sh,
+                      int64_t& sc, int64_t& sl, int64_t& NC, int64_t& LC) {
+        TORCH_CHECK(t.scalar_type() == at::kBFloat16);
+        ptr = reinterpret_cast<const uint16_t*>(t.data_ptr());
+        sb = t.stride(0);
+        sh = t.stride(1);
+        if (t.dim() == 4) {
+            NC = 1;
+            LC = t.size(2);
+            sc = 0;
+            sl = t.stride(2);
+        } else {
+            TORCH_CHECK(t.dim() == 5);
+            NC = t.size(2);
+            LC = t.size(3);
+            sc = t.stride(2);
+            sl = t.stride(3);
+        }
+        TORCH_CHECK(sl % 8 == 0 && sc % 8 == 0 && sh % 8 == 0 && sb % 8 == 0,
+                    "KV strides must be 16B-aligned (multiples of 8 elements)");
+    };
+    int64_t nc2, lc2;
+    set_kv(k, p.k, p.k_sb, p.k_sh, p.k_sc, p.k_sl, p.NC, p.LC);
+    set_kv(v, p.v, p.v_sb, p.v_sh, p.v_sc, p.v_sl, nc2, lc2);
+    TORCH_CHECK(nc2 == p.NC && lc2 == p.LC, "k/v shape mismatch");
+    TORCH_CHECK(p.q_sl % 8 == 0 && p.q_sb % 8 == 0 && p.q_sh % 8 == 0);
+
+    auto o = at::empty({(long)B, (long)Lq, (long)H, 64L}, q.options());
+    p.o = reinterpret_cast<uint16_t*>(o.data_ptr());
+    launch_flash_attention_d64(p, cur_stream());
+    return o.permute({0, 2, 1, 3});  // logical [B,H,Lq,64]
+}
+
+std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b);
+
+}  // namespace
+
+// defined in attention.hip
+void launch_mfma_probe(const float* a, const float* b, float* d, hipStream_t stream);
+
+namespace {
+std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b) {
+    TORCH_CHECK(a.is_cuda() && a.sizes() == (at::IntArrayRef{64, 8}));
+    auto af = a.to(at::kFloat).contiguous(), bf = b.to(at::kFloat).contiguous();
+    auto d = at::zeros({64, 4}, af.options());
+    launch_mfma_probe(af.data_ptr<float>(), bf.data_ptr<float>(), d.data_ptr<float>(),
+                      cur_stream());
+    return {d};
+}
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("group_norm_stats", &group_norm_stats, "per-group moments [2,N,G,1,1,1]");
+    m.def("group_norm_apply", &group_norm_apply, "normalize+affine(+SiLU) from given moments");
+    m.def("group_norm_silu", &group_norm_silu, "fused GroupNorm(+SiLU)");
+    m.def("geglu", &geglu, "a * gelu(gate) over last-dim halves");
+    m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
+    m.def("mfma_probe", &mfma_probe, "dump mfma_f32_16x16x32_bf16 fragment mapping");
+}

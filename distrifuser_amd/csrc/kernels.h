@@ -1,0 +1,42 @@
+// Host-side launcher declarations (implemented in the .hip translation units).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+enum DfaDtype : int { DFA_BF16 = 0, DFA_F16 = 1, DFA_F32 = 2 };
+
+// ---- GroupNorm family ------------------------------------------------------
+// partial: fp32 [2, N*G] zero-initialized; accumulates sum and sumsq.
+void launch_gn_stats_partial(const void* x, float* partial, int64_t group_len,
+                             int n_groups_total, int dtype, hipStream_t stream);
+// out: [2, N*G] in `dtype`; mean = partial/count.
+void launch_gn_finalize(const float* partial, void* out, float inv_count,
+                        int n_groups_total, int dtype, hipStream_t stream);
+// normalize+affine(+SiLU) with externally supplied fp32 moments [N*G] each.
+void launch_gn_apply(const void* x, void* y, const float* mean, const float* meansq,
+                     const void* weight, const void* bias, float eps, int64_t hw,
+                     int C, int G, int N, bool silu, int dtype, hipStream_t stream);
+
+// ---- GEGLU ------------------------------------------------------------------
+// in: [rows, 2*inner]; out: [rows, inner]; out = a * gelu(gate).
+void launch_geglu(const void* in, void* out, int64_t rows, int64_t inner, int dtype,
+                  hipStream_t stream);
+
+// ---- Flash attention (bf16, head_dim 64) ------------------------------------
+// q: logical [B, H, Lq, 64]; k/v: logical [B, H, NC, LC, 64] (NC stale-KV
+// chunks of LC tokens; NC=1 for plain attention). All strides in ELEMENTS,
+// innermost head_dim contiguous. o: [B, Lq, H, 64] contiguous output.
+struct FlashAttnParams {
+    const uint16_t* q;
+    const uint16_t* k;
+    const uint16_t* v;
+    uint16_t* o;
+    int B, H;
+    int64_t Lq, NC, LC;  // Lkv = NC * LC
+    int64_t q_sb, q_sh, q_sl;
+    int64_t k_sb, k_sh, k_sc, k_sl;
+    int64_t v_sb, v_sh, v_sc, v_sl;
+    float scale;  // 1/sqrt(64)
+};
+void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream);

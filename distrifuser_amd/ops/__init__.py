@@ -9,6 +9,7 @@ Set DFA_FORCE_EAGER=1 to force the eager path on GPU (A/B debugging only).
 
 from .dispatch import (
     flash_attention,
+    flash_attention_chunked,
     geglu,
     group_norm_apply,
     group_norm_silu,
@@ -19,6 +20,7 @@ from .dispatch import (
 
 __all__ = [
     "flash_attention",
+    "flash_attention_chunked",
     "geglu",
     "group_norm_apply",
     "group_norm_silu",

@@ -52,19 +52,54 @@ def _use_hip(x: torch.Tensor) -> bool:
 # -- public ops --------------------------------------------------------------
 
 
-def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
-    """Attention over (possibly stale, strided) full-sequence KV.
+def _flash_ok(q: torch.Tensor, *kv: torch.Tensor) -> bool:
+    if q.dtype != torch.bfloat16 or q.shape[-1] != 64:
+        return False
+    for t in (q, *kv):
+        if t.stride(-1) != 1:
+            return False
+        if any(s % 8 != 0 for s in t.stride()[:-1]):
+            return False
+    return True
 
-    q: [B, H, Lq, D]; k, v: [B, H, Lkv, D] (arbitrary stride in the Lkv dim
-    so the stale-KV flat comm buffer can be consumed without a torch.cat).
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """Attention over (possibly strided) full-sequence KV.
+
+    q: [B, H, Lq, D]; k, v: [B, H, Lkv, D] (arbitrary stride in the Lkv dim).
     """
     if _use_hip(q):
-        ext = hip_ext()
-        if q.dtype == torch.bfloat16 and q.shape[-1] in (64,) and hasattr(ext, "flash_attention"):
-            return ext.flash_attention(q, k, v)
+        if _flash_ok(q, k, v):
+            return hip_ext().flash_attention(q, k, v)
         # Non-bf16 / odd head-dims ride PyTorch's SDPA (rocm AOTriton path).
         return eager.flash_attention(q, k, v)
     return eager.flash_attention(q, k, v)
+
+
+def flash_attention_chunked(
+    q: torch.Tensor, kv_chunks: torch.Tensor, heads: int, dim_head: int
+) -> torch.Tensor:
+    """Displaced-patch attention: q [B, Lq, inner]; kv_chunks is the flat comm
+    buffer viewed as [n_peers, B, L_local, 2*inner] (strided — peer rows live
+    in the flat buffer). On GPU the gfx950 kernel walks the chunks in place;
+    the eager path materializes the concatenated KV.
+    """
+    b, lq, inner = q.shape
+    n, _, l, _ = kv_chunks.shape
+    if _use_hip(q):
+        q4 = q.view(b, lq, heads, dim_head).permute(0, 2, 1, 3)
+        k = kv_chunks[..., :inner].unflatten(-1, (heads, dim_head)).permute(1, 3, 0, 2, 4)
+        v = kv_chunks[..., inner:].unflatten(-1, (heads, dim_head)).permute(1, 3, 0, 2, 4)
+        if _flash_ok(q4, k, v):
+            o = hip_ext().flash_attention(q4, k, v)  # [B,H,Lq,D] (view of B,Lq,H,D)
+            return o.transpose(1, 2).reshape(b, lq, inner)
+    full_kv = kv_chunks.permute(1, 0, 2, 3).reshape(b, n * l, 2 * inner)
+    k, v = full_kv.split(inner, dim=-1)
+    q4 = q.view(b, lq, heads, dim_head).transpose(1, 2)
+    k = k.view(b, n * l, heads, dim_head).transpose(1, 2)
+    v = v.view(b, n * l, heads, dim_head).transpose(1, 2)
+    out = flash_attention(q4, k.contiguous(), v.contiguous())
+    return out.transpose(1, 2).reshape(b, lq, inner)
 
 
 def group_norm_stats(x: torch.Tensor, num_groups: int) -> torch.Tensor:

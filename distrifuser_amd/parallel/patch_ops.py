@@ -358,12 +358,11 @@ class PatchSelfAttention(nn.Module):
             if cfg.mode != "no_sync":
                 comm.enqueue(self._idx)  # fresh slice already staged in-slot
 
-        # [n, B, L, 2C] -> [B, n*L, 2C]; peers are ordered by patch index so
-        # the concatenated sequence is the full image in row order.
-        full_kv = self._buffer_view.permute(1, 0, 2, 3).reshape(
-            b, cfg.n_device_per_batch * l, self.to_kv.out_features
-        )
-        return self.to_out(self._attention(q, full_kv))
+        # The buffer view [n, B, L, 2C] IS the full displaced KV (peers are
+        # ordered by patch index = full image row order); the kernel walks the
+        # peer chunks in place — no torch.cat on the hot path.
+        out = ops.flash_attention_chunked(q, self._buffer_view, self.heads, self.dim_head)
+        return self.to_out(out)
 
 
 class CachedCrossAttention(nn.Module):

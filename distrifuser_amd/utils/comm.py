@@ -68,6 +68,9 @@ class PatchParallelismCommManager:
         numel = 1
         for d in shape:
             numel *= int(d)
+        # 16B-align every slot start so the gfx950 kernels can issue uint4
+        # loads straight out of the flat buffer (8 elements at 2B dtypes).
+        self.numels = (self.numels + 7) // 8 * 8
         self.starts.append(self.numels)
         self.ends.append(self.numels + numel)
         self.shapes.append(torch.Size(shape))
@@ -83,6 +86,9 @@ class PatchParallelismCommManager:
         if self.numels == 0:
             self.handles = []
             return
+        # row stride (= per-peer chunk stride seen by the attention kernel)
+        # must stay 16B-aligned too
+        self.numels = (self.numels + 7) // 8 * 8
         self.buffer = torch.empty(
             (n, self.numels), dtype=self.torch_dtype, device=self.distri_config.device
         )

@@ -13,7 +13,10 @@ def _roundtrip(rank, world_size):
     i0 = comm.register_tensor((2, 3), torch.float32, layer_type="attn")
     i1 = comm.register_tensor((4,), torch.float32, layer_type="gn")
     comm.create_buffer()
-    assert comm.numels == 10
+    # slot starts and the row stride are 16B-aligned (8-element grid):
+    # (2,3) at [0,6), (4,) at [8,12), row stride padded to 16
+    assert comm.starts == [0, 8]
+    assert comm.numels == 16
 
     t0 = torch.full((2, 3), float(rank))
     t1 = torch.arange(4, dtype=torch.float32) + rank * 10

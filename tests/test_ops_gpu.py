@@ -1,0 +1,165 @@
+"""GPU numerics: every gfx950 kernel vs the plain-PyTorch fp32 reference
+(ops/eager.py). Run on an MI355X box: pytest tests -m gpu -x -q"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from distrifuser_amd import ops
+    from distrifuser_amd.ops import eager
+    from distrifuser_amd.ops.dispatch import hip_ext
+else:  # collected on CPU boxes but never run
+    ops = eager = hip_ext = None
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs ROCm GPU")
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+@requires_gpu
+def test_extension_loads():
+    ext = hip_ext()
+    for fn in ("group_norm_stats", "group_norm_apply", "group_norm_silu", "geglu",
+               "flash_attention", "mfma_probe"):
+        assert hasattr(ext, fn)
+
+
+@requires_gpu
+def test_mfma_fragment_layout():
+    """Verify the A/B fragment maps assumed by attention.hip:
+    A[row=l&15][k=(l>>4)*8+j], B[k=(l>>4)*8+j][col=l&15],
+    D[row=(l>>4)*4+r][col=l&15]."""
+    torch.manual_seed(0)
+    # values exactly representable in bf16
+    a_frag = torch.randint(-8, 8, (64, 8), dtype=torch.float32, device=_dev())
+    b_frag = torch.randint(-8, 8, (64, 8), dtype=torch.float32, device=_dev())
+    (d,) = hip_ext().mfma_probe(a_frag, b_frag)
+    d = d.cpu()
+
+    A = torch.zeros(16, 32)
+    B = torch.zeros(32, 16)
+    for lane in range(64):
+        lo, hi = lane & 15, lane >> 4
+        for j in range(8):
+            A[lo, hi * 8 + j] = a_frag[lane, j]
+            B[hi * 8 + j, lo] = b_frag[lane, j]
+    D = A @ B
+    for lane in range(64):
+        lo, hi = lane & 15, lane >> 4
+        for r in range(4):
+            assert abs(d[lane, r].item() - D[hi * 4 + r, lo].item()) < 1e-3, (
+                f"lane {lane} r {r}: got {d[lane, r]}, want {D[hi * 4 + r, lo]}"
+            )
+
+
+def _attn_ref(q, k, v):
+    return eager.flash_attention(q.float(), k.float(), v.float())
+
+
+@requires_gpu
+@pytest.mark.parametrize(
+    "b,h,lq,lkv",
+    [(1, 2, 64, 64), (1, 4, 100, 200), (2, 10, 1024, 1024), (1, 5, 4096, 4096), (1, 2, 37, 77)],
+)
+def test_flash_attention_vs_fp32(b, h, lq, lkv):
+    torch.manual_seed(0)
+    dev = _dev()
+    q = torch.randn(b, h, lq, 64, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(b, h, lkv, 64, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(b, h, lkv, 64, device=dev, dtype=torch.bfloat16)
+    out = ops.flash_attention(q, k, v).float()
+    ref = _attn_ref(q, k, v)
+    err = (out - ref).abs().max().item()
+    assert err < 0.03, f"max err {err}"
+
+
+@requires_gpu
+def test_flash_attention_chunked_matches_cat():
+    """5-D chunked KV (the stale-KV flat-buffer layout) == concatenated KV."""
+    torch.manual_seed(0)
+    dev = _dev()
+    b, heads, dim_head, l, n = 1, 4, 64, 256, 4
+    inner = heads * dim_head
+    q = torch.randn(b, l, inner, device=dev, dtype=torch.bfloat16)
+    # emulate the flat buffer: [n, numel_slot + pad] rows
+    slot = b * l * 2 * inner
+    buf = torch.randn(n, slot + 64, device=dev, dtype=torch.bfloat16)
+    kv_chunks = buf[:, :slot].view(n, b, l, 2 * inner)
+    out = ops.flash_attention_chunked(q, kv_chunks, heads, dim_head).float()
+
+    full = kv_chunks.permute(1, 0, 2, 3).reshape(b, n * l, 2 * inner).float()
+    kf, vf = full.split(inner, dim=-1)
+    qf = q.float().view(b, l, heads, dim_head).transpose(1, 2)
+    kf = kf.view(b, n * l, heads, dim_head).transpose(1, 2)
+    vf = vf.view(b, n * l, heads, dim_head).transpose(1, 2)
+    ref = eager.flash_attention(qf, kf, vf).transpose(1, 2).reshape(b, l, inner)
+    err = (out - ref).abs().max().item()
+    assert err < 0.03, f"max err {err}"
+
+
+@requires_gpu
+def test_group_norm_stats_gpu():
+    torch.manual_seed(0)
+    x = torch.randn(2, 64, 33, 40, device=_dev(), dtype=torch.bfloat16)
+    out = ops.group_norm_stats(x, 32).float()
+    ref = eager.group_norm_stats(x.float().cpu(), 32)
+    assert out.shape == (2, 2, 32, 1, 1, 1)
+    assert (out.cpu() - ref).abs().max() < 5e-3
+
+
+@requires_gpu
+def test_group_norm_apply_gpu():
+    torch.manual_seed(0)
+    dev = _dev()
+    x = torch.randn(2, 64, 16, 24, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(64, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(64, device=dev, dtype=torch.bfloat16)
+    stats = ops.group_norm_stats(x, 8)
+    for silu in (False, True):
+        out = ops.group_norm_apply(x, stats[0], stats[1], w, b, 1e-5, silu=silu).float()
+        ref = eager.group_norm_apply(
+            x.float().cpu(), stats[0].float().cpu(), stats[1].float().cpu(),
+            w.float().cpu(), b.float().cpu(), 1e-5, silu=silu,
+        )
+        assert (out.cpu() - ref).abs().max() < 0.05
+
+
+@requires_gpu
+def test_group_norm_silu_fused_gpu():
+    torch.manual_seed(0)
+    dev = _dev()
+    for shape, g in [((1, 320, 64, 64), 32), ((2, 64, 15, 30), 8), ((1, 8, 3, 3), 4)]:
+        x = torch.randn(*shape, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(shape[1], device=dev, dtype=torch.bfloat16)
+        b = torch.randn(shape[1], device=dev, dtype=torch.bfloat16)
+        out = ops.group_norm_silu(x, g, w, b, 1e-6, silu=True).float()
+        ref = eager.group_norm_silu(x.float().cpu(), g, w.float().cpu(), b.float().cpu(),
+                                    1e-6, silu=True)
+        err = (out.cpu() - ref).abs().max().item()
+        assert err < 0.05, f"{shape} g={g}: max err {err}"
+
+
+@requires_gpu
+def test_geglu_gpu():
+    torch.manual_seed(0)
+    for shape in [(2, 128, 256), (1, 77, 48), (3, 5, 10)]:
+        h = torch.randn(*shape, device=_dev(), dtype=torch.bfloat16)
+        out = ops.geglu(h).float()
+        ref = eager.geglu(h.float().cpu())
+        assert (out.cpu() - ref).abs().max() < 0.02, shape
+
+
+@requires_gpu
+def test_gpu_ops_fail_loudly_without_ext(monkeypatch):
+    """On a GPU box, a missing extension must raise, not fall back silently."""
+    from distrifuser_amd.ops import dispatch
+
+    monkeypatch.setattr(dispatch, "_EXT", None)
+    monkeypatch.setattr(dispatch, "_EXT_ERR", ImportError("simulated"))
+    x = torch.randn(1, 8, 4, 4, device=_dev(), dtype=torch.bfloat16)
+    with pytest.raises(RuntimeError, match="extension"):
+        dispatch.group_norm_silu(x, 2, None, None, 1e-5)
