@@ -326,7 +326,9 @@ class UNet2DConditionNative(nn.Module):
         self.up_blocks = nn.ModuleList(up_blocks)
 
         self.conv_norm_out = factory.group_norm(groups, ch0, eps=1e-5, fuse_silu=True)
-        self.conv_out = factory.conv2d(ch0, cfg.out_channels, 3, 1, 1)
+        # conv_out is input-channel-sharded under TP (reference
+        # distri_sdxl_unet_tp.py:27-38); conv_in stays replicated (4 inputs)
+        self.conv_out = factory.conv2d(ch0, cfg.out_channels, 3, 1, 1, tp_shard=True)
 
     def forward(
         self,

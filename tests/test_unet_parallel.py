@@ -166,6 +166,48 @@ def test_tensor_parallel_runs_ws2():
     assert torch.isfinite(out[0]).all()
 
 
+@torch.no_grad()
+def _tp_loaded_worker(rank, world_size, sd_bytes):
+    import io
+
+    cfg = DistriConfig(
+        height=H * 8, width=W * 8, do_classifier_free_guidance=False,
+        parallelism="tensor", use_cuda_graph=False, device="cpu",
+    )
+    torch.manual_seed(7)
+    unet = DistriUNet(TINY_UNET, cfg).eval()
+    sd = torch.load(io.BytesIO(sd_bytes), weights_only=False)
+    from distrifuser_amd.models.weights import load_unet_checkpoint
+
+    load_unet_checkpoint(unet.unet, sd)
+    sample, ehs, added = _tiny_inputs(1)
+    unet.set_counter(0)
+    return unet(sample, 3.0, ehs, added).clone()
+
+
+def test_tensor_parallel_loads_full_checkpoint_ws2():
+    """TP shards sliced from a full diffusers-layout dict reproduce the
+    single-process model's output."""
+    import io
+
+    from distrifuser_amd.models.weights import export_diffusers_state_dict
+
+    cfg, unet = _build_unet(parallelism="patch", do_cfg=False)
+    sd = export_diffusers_state_dict(unet.unet)
+    buf = io.BytesIO()
+    torch.save(sd, buf)
+
+    sample, ehs, added = _tiny_inputs(1)
+    unet.set_counter(0)
+    with torch.no_grad():
+        ref = unet(sample, 3.0, ehs, added)
+
+    out = run_distributed(2, _tp_loaded_worker, (buf.getvalue(),))
+    for r in (0, 1):
+        err = (out[r] - ref).abs().max()
+        assert err < 5e-4, f"rank {r}: max err {err}"
+
+
 def test_tensor_parallel_cfg_split_ws2():
     """Exercises the CFG-pair exchange the reference left broken
     (reference distri_sdxl_unet_tp.py:159-162)."""
