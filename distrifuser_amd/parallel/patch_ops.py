@@ -125,26 +125,43 @@ class PatchConv2d(nn.Module):
             dist.all_gather(self._buffer_list, boundary, group=cfg.batch_group)
         split = cfg.split_idx()
         n = cfg.n_device_per_batch
-        parts = []
-        if split == 0:
-            parts.append(x.new_zeros(x.shape[0], x.shape[1], halo, x.shape[3]))
-        else:
-            parts.append(self._buffer_list[split - 1][1])  # neighbour-above's bottom rows
-        parts.append(x)
-        if split == n - 1:
-            parts.append(x.new_zeros(x.shape[0], x.shape[1], halo, x.shape[3]))
-        else:
-            parts.append(self._buffer_list[split + 1][0])  # neighbour-below's top rows
-        padded = torch.cat(parts, dim=2)
-        out = F.conv2d(
-            padded,
-            self.conv.weight,
-            self.conv.bias,
-            stride=self.conv.stride[0],
-            padding=(0, self.conv.padding[1]),
-        )
+        top = None if split == 0 else self._buffer_list[split - 1][1]
+        bot = None if split == n - 1 else self._buffer_list[split + 1][0]
+        out = self._conv_with_halos(x, top, bot)
         if not self.state.use_sync_comm and cfg.mode != "no_sync":
             comm.enqueue(self._idx, boundary)
+        return out
+
+    def _conv_with_halos(self, x, top, bot):
+        """Convolve the local band with neighbour halo rows WITHOUT
+        materializing cat([halo, x, halo]) (the reference copies the whole
+        input per conv, pp/conv2d.py:72-88): run the conv zero-padded on x
+        and recompute only the boundary output rows from the halos.
+
+        Specialized for the SD-family k=3/pad=1 convs; other shapes take the
+        concat fallback.
+        """
+        conv = self.conv
+        k = conv.kernel_size[0]
+        s = conv.stride[0]
+        pw = conv.padding[1]
+        if k != 3 or conv.padding[0] != 1 or s not in (1, 2):
+            parts = [p for p in (top, x, bot) if p is not None]
+            pad_top = conv.padding[0] if top is None else 0
+            pad_bot = conv.padding[0] if bot is None else 0
+            padded = torch.cat(parts, dim=2) if len(parts) > 1 else parts[0]
+            padded = F.pad(padded, [0, 0, pad_top, pad_bot])
+            return F.conv2d(padded, conv.weight, conv.bias, stride=s, padding=(0, pw))
+
+        out = F.conv2d(x, conv.weight, conv.bias, stride=s, padding=(1, pw))
+        # first output row: window covers rows [-1, 1] -> needs the top halo
+        if top is not None:
+            tin = torch.cat([top, x[:, :, :2]], dim=2)
+            out[:, :, :1] = F.conv2d(tin, conv.weight, conv.bias, stride=s, padding=(0, pw))
+        # last output row (stride 1 only; stride 2's last window ends inside x)
+        if s == 1 and bot is not None:
+            bin_ = torch.cat([x[:, :, -2:], bot], dim=2)
+            out[:, :, -1:] = F.conv2d(bin_, conv.weight, conv.bias, stride=s, padding=(0, pw))
         return out
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
