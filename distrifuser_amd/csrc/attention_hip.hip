@@ -2,90 +2,96 @@
 // Flash-attention forward, gfx950, bf16, head_dim 64 (K1/K2/K8 of SURVEY
 // §2.4a): rectangular local-query x global-(stale)-KV attention.
 //
-// Design (cdna_hip_programming.md §B "fused attention prefill" ladder):
-// * one 4-wave workgroup per (batch, head, 64-row Q tile); each wave owns a
-//   16-row Q sub-tile, Q held in registers (2 MFMA A-fragments per lane).
-// * KV streamed in 64-token LDS tiles shared by the 4 waves; K row-major
-//   [64][64], V stored TRANSPOSED [64 d][64 t] so both QK^T and PV read
-//   contiguous 16B B-fragments (ds_read_b128).
-// * LDS XOR swizzle byte ^= ((row & 7) << 4) on every row-strided buffer —
-//   row-major [.][64] bf16 tiles are a 16-way bank conflict otherwise
-//   (guide §6 Guideline 4).
-// * online softmax entirely wave-parallel: the MFMA C-layout puts one kv
-//   column per lane (col = lane&15), so row max/sum are __shfl_xor(1,2,4,8)
-//   reductions — no serial-lane softmax (guide common-mistake #6).
+// v2 structure (cdna_hip_programming.md §B "8-warp 32x32 ladder"):
+// * 4-wave workgroup per (batch, head, 128-row Q tile); each wave owns 32 Q
+//   rows held in registers (4 k-slice A/B fragments).
+// * KV streamed in 64-token LDS tiles shared by the waves; K row-major
+//   [64][64], V TRANSPOSED [d][t]; XOR swizzle byte^=((row&7)<<4) breaks the
+//   16-way bank conflict of row-major [.][64] bf16 tiles (guide §6 G4).
+// * SWAPPED QK^T: S^T = K_tile x Q^T via mfma_f32_32x32x16_bf16, so each
+//   lane holds 16 score rows of ONE q column (q = lane&31) — the online-
+//   softmax max/sum are 15 in-register ops + ONE shfl_xor(32) with the
+//   partner lane, instead of 30+ cross-lane shuffles per tile (guide
+//   common-mistake #6 / §B "swapped QK^T").
+// * P stays in registers: v_cvt_pk_bf16_f32 packs + shfl_xor(32) partner
+//   exchange build the PV B-fragments directly (guide T12), no P LDS
+//   round-trip.
+// * PV is also swapped: O^T = V^T x P^T accumulates in 2x16 AGPRs per lane.
 // * stale-KV chunking: KV tokens come from NC flat-comm-buffer chunks of LC
-//   tokens (k_sc/v_sc chunk strides), so the displaced-patch KV is consumed
-//   in place with zero torch.cat.
+//   tokens (k_sc/v_sc strides) — the displaced-patch KV is consumed in place
+//   with zero torch.cat.
 //
-// MFMA v_mfma_f32_16x16x32_bf16 fragment maps (A/B assumed per CDNA ISA, C/D
-// verified in the guide §3): A: [row=l&15][k=(l>>4)*8+j]; B: [k=(l>>4)*8+j]
-// [col=l&15]; C/D: [row=(l>>4)*4+r][col=l&15]. tests/test_ops_gpu.py checks
-// the whole kernel against fp32 SDPA.
+// Fragment maps (verified on MI355X by tests/test_ops_gpu.py probes):
+//   16x16x32: A[row=l&15][k=(l>>4)*8+j]  B[k][col=l&15]  D[row=(l>>4)*4+r][col=l&15]
+//   32x32x16: A[row=l&31][k=(l>>5)*8+j]  B[k][col=l&31]
+//             D[row=(r&3)+8*(r>>2)+4*(l>>5)][col=l&31]
 
 #include "common.h"
 #include "kernels.h"
 
 namespace {
 
-constexpr int QB = 16;     // q rows per wave
+constexpr int QW = 32;     // q rows per wave
 constexpr int WAVES = 4;   // waves per block
-constexpr int QBLK = QB * WAVES;
+constexpr int QBLK = QW * WAVES;  // 128
 constexpr int KVB = 64;    // kv tokens per LDS tile
 constexpr int D = 64;
+
+typedef float float4v_ __attribute__((ext_vector_type(4)));
+typedef float float16v __attribute__((ext_vector_type(16)));
 
 __device__ __forceinline__ int swz(int row, int byte_off) {
     return byte_off ^ ((row & 7) << 4);
 }
 
-typedef float float4v_ __attribute__((ext_vector_type(4)));
+__device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
+    uint32_t r;
+    asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+    return r;
+}
 
 __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(FlashAttnParams p) {
-    __shared__ char k_lds[KVB * D * 2];       // [t][d] bf16, swizzled rows
-    __shared__ char vt_lds[D * KVB * 2];      // [d][t] bf16, swizzled rows
-    __shared__ char p_lds[WAVES][QB * KVB * 2];  // per-wave P tile [q][t]
+    __shared__ char k_lds[KVB * D * 2];   // [t][d] bf16, swizzled rows
+    __shared__ char vt_lds[D * KVB * 2];  // [d][t] bf16, swizzled rows
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
     const int lane = tid % WAVE_SIZE;
-    const int lo = lane & 15;
-    const int hi = lane >> 4;
+    const int lo = lane & 31;   // q column of this lane
+    const int hi = lane >> 5;   // partner-half index
 
-    const int bh = blockIdx.y;  // b * H + h
+    const int bh = blockIdx.y;
     const int b = bh / p.H;
     const int h = bh % p.H;
-    const int64_t q0 = (int64_t)blockIdx.x * QBLK;  // first q row of the block
+    const int64_t q0 = (int64_t)blockIdx.x * QBLK;
 
     const int64_t Lkv = p.NC * p.LC;
     const uint16_t* qbase = p.q + b * p.q_sb + h * p.q_sh;
     const uint16_t* kbase = p.k + b * p.k_sb + h * p.k_sh;
     const uint16_t* vbase = p.v + b * p.v_sb + h * p.v_sh;
 
-    // ---- load Q fragments (row = q0 + wave*16 + lo; d = hi*8 + 32*ks) ----
-    short8 qfrag[2];
-    const int64_t qrow = q0 + wave * QB + lo;
+    // ---- Q fragments: qf[ks] = Q[q = q0 + wave*32 + lo][d = ks*16 + hi*8 ..] ----
+    short8 qf[4];
+    const int64_t qrow = q0 + wave * QW + lo;
     const bool qvalid = qrow < p.Lq;
     {
         const uint16_t* qp = qbase + (qvalid ? qrow : (p.Lq - 1)) * p.q_sl;
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
-            qfrag[ks] = *reinterpret_cast<const short8*>(qp + hi * 8 + 32 * ks);
+        for (int ks = 0; ks < 4; ++ks)
+            qf[ks] = *reinterpret_cast<const short8*>(qp + ks * 16 + hi * 8);
     }
 
-    // online-softmax state: this lane participates in rows hi*4 + r
-    float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
-    float l_run[4] = {0.f, 0.f, 0.f, 0.f};
-    float4v_ oacc[4];  // [dblk] -> D[row=(hi*4+r)][dcol = dblk*16 + lo]
-#pragma unroll
-    for (int i = 0; i < 4; ++i) oacc[i] = {0.f, 0.f, 0.f, 0.f};
+    float m_run = -1e30f;
+    float l_run = 0.f;
+    float16v ot[2] = {};  // O^T tiles: [dt] -> rows d = dt*32 + crow(r,hi), col q=lo
 
     const int n_tiles = (int)((Lkv + KVB - 1) / KVB);
     for (int tile = 0; tile < n_tiles; ++tile) {
         const int64_t t0 = (int64_t)tile * KVB;
-        // ---- stage K [t][d] and V^T [d][t] (256 threads cooperative) ----
+        // ---- cooperative staging: K [t][d], V^T [d][t] ----
         {
-            const int tl = tid / 8;        // 0..31: local token row pair index
-            const int d8 = tid % 8;        // 16B column
+            const int tl = tid / 8;
+            const int d8 = tid % 8;
 #pragma unroll
             for (int rep = 0; rep < 2; ++rep) {
                 const int t_local = tl + rep * 32;
@@ -110,107 +116,110 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
         }
         __syncthreads();
 
-        // ---- S = scale * Q K^T over 4 16-col blocks ----
-        // A = Q tile: lane reads Q[row=l&15][d=(l>>4)*8+j] (qfrag).
-        // B = K^T:    lane reads K^T[d=(l>>4)*8+j][t=l&15] = K[t][d] (kfrag;
-        //             same lane arithmetic as A, t from the 16-col block).
-        // D[row=q][col=t] lands at [row=(l>>4)*4+r][col=l&15].
-        float4v_ s[4];
 #pragma unroll
-        for (int blk = 0; blk < 4; ++blk) {
-            float4v_ acc = {0.f, 0.f, 0.f, 0.f};
+        for (int st = 0; st < 2; ++st) {  // two 32-token sub-tiles
+            // ---- S^T[kv32][q32] = K_sub x Q^T ----
+            float16v s = {};
 #pragma unroll
-            for (int ks = 0; ks < 2; ++ks) {
-                const int t = blk * 16 + lo;
+            for (int ks = 0; ks < 4; ++ks) {
+                const int t = st * 32 + lo;
                 short8 kfrag = *reinterpret_cast<const short8*>(
-                    &k_lds[t * 128 + swz(t, (hi * 8 + 32 * ks) * 2)]);
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], kfrag, acc, 0, 0, 0);
+                    &k_lds[t * 128 + swz(t, (ks * 16 + hi * 8) * 2)]);
+                s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
             }
-            s[blk] = acc;
-        }
+            // lane holds S^T rows crow(r) = (r&3)+8*(r>>2)+4*hi for q col lo
+            float tm = -1e30f;
+            float pv[16];
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const int64_t tglob = t0 + st * 32 + crow;
+                float v = (tglob < Lkv) ? s[r] * p.scale : -1e30f;
+                pv[r] = v;
+                tm = fmaxf(tm, v);
+            }
+            tm = fmaxf(tm, __shfl_xor(tm, 32, 64));  // partner holds the other 16 rows
+            const float m_new = fmaxf(m_run, tm);
+            const float corr = __expf(m_run - m_new);
+            m_run = m_new;
+            l_run *= corr;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                ot[0][r] *= corr;
+                ot[1][r] *= corr;
+            }
+            float tsum = 0.f;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                pv[r] = __expf(pv[r] - m_new);
+                tsum += pv[r];
+            }
+            tsum += __shfl_xor(tsum, 32, 64);
+            l_run += tsum;
 
-        // ---- masking + online softmax (rows hi*4+r, col lo per blk) ----
-        float pmax[4];
+            // ---- pack P to bf16 B-fragments (guide T12: cvt_pk + partner
+            // exchange; P^T[t][q] needs rows hi*8..hi*8+7 per k-slice) ----
+            uint32_t w[8], wp[8];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) pmax[r] = -1e30f;
-#pragma unroll
-        for (int blk = 0; blk < 4; ++blk) {
-            const int64_t col = t0 + blk * 16 + lo;
-            const bool valid = col < Lkv;
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                float v = valid ? s[blk][r] * p.scale : -1e30f;
-                s[blk][r] = v;
-                pmax[r] = fmaxf(pmax[r], v);
+            for (int i = 0; i < 8; ++i) {
+                w[i] = cvt_pk_bf16(pv[2 * i], pv[2 * i + 1]);
+                wp[i] = __shfl_xor(w[i], 32, 64);
             }
-        }
-        // row reduce across the 16 lanes of the row group
+            short8 pb[2];
 #pragma unroll
-        for (int off = 1; off < 16; off <<= 1)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) pmax[r] = fmaxf(pmax[r], __shfl_xor(pmax[r], off, 64));
+            for (int kt = 0; kt < 2; ++kt) {
+                // B frag j=0..7 -> P^T rows kt*16 + hi*8 + j.
+                // own regs r cover rows (r&3)+8*(r>>2)+4*hi; packed word i
+                // holds regs {2i, 2i+1} = rows {(2i&3)+8*(2i>>2)+4*hi, ...}.
+                // For hi=0: rows kt*16+0..7 = own words {4kt,4kt+1} (rows
+                // kt*16+0..3) then partner words {4kt,4kt+1} (rows kt*16+4..7).
+                // For hi=1: rows kt*16+8..15 = partner {4kt+2,4kt+3} then own
+                // {4kt+2,4kt+3}.
+                uint32_t b0, b1, b2, b3;
+                if (hi == 0) {
+                    b0 = w[4 * kt + 0];
+                    b1 = w[4 * kt + 1];
+                    b2 = wp[4 * kt + 0];
+                    b3 = wp[4 * kt + 1];
+                } else {
+                    b0 = wp[4 * kt + 2];
+                    b1 = wp[4 * kt + 3];
+                    b2 = w[4 * kt + 2];
+                    b3 = w[4 * kt + 3];
+                }
+                uint32_t* pbw = reinterpret_cast<uint32_t*>(&pb[kt]);
+                pbw[0] = b0;
+                pbw[1] = b1;
+                pbw[2] = b2;
+                pbw[3] = b3;
+            }
 
-        float lsum[4];
+            // ---- O^T += V^T x P^T ----
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const float m_new = fmaxf(m_run[r], pmax[r]);
-            const float corr = __expf(m_run[r] - m_new);
-            m_run[r] = m_new;
-            l_run[r] *= corr;
+            for (int dt = 0; dt < 2; ++dt) {
 #pragma unroll
-            for (int dblk = 0; dblk < 4; ++dblk) oacc[dblk][r] *= corr;
-            lsum[r] = 0.f;
-        }
-        // P = exp(S - m); write bf16 P tile to this wave's LDS
-#pragma unroll
-        for (int blk = 0; blk < 4; ++blk) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const float e = __expf(s[blk][r] - m_run[r]);
-                lsum[r] += e;
-                const int prow = hi * 4 + r;
-                const int pcol = blk * 16 + lo;
-                *reinterpret_cast<uint16_t*>(
-                    &p_lds[wave][prow * 128 + swz(prow, pcol * 2)]) =
-                    __builtin_bit_cast(uint16_t, __float2bfloat16(e));
+                for (int kt = 0; kt < 2; ++kt) {
+                    const int d = dt * 32 + lo;
+                    short8 vf = *reinterpret_cast<const short8*>(
+                        &vt_lds[d * 128 + swz(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
+                    ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
+                }
             }
         }
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) lsum[r] += __shfl_xor(lsum[r], off, 64);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) l_run[r] += lsum[r];
-
-        // ---- O += P V : A = P[q=lo][t=hi*8+32ks+j], B = V^T[d=lo+16dblk][t] ----
-#pragma unroll
-        for (int dblk = 0; dblk < 4; ++dblk) {
-#pragma unroll
-            for (int ks = 0; ks < 2; ++ks) {
-                const int prow = lo;
-                short8 pfrag = *reinterpret_cast<const short8*>(
-                    &p_lds[wave][prow * 128 + swz(prow, (hi * 8 + 32 * ks) * 2)]);
-                const int vrow = dblk * 16 + lo;
-                short8 vfrag = *reinterpret_cast<const short8*>(
-                    &vt_lds[vrow * 128 + swz(vrow, (hi * 8 + 32 * ks) * 2)]);
-                oacc[dblk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag, oacc[dblk], 0, 0, 0);
-            }
-        }
-        __syncthreads();  // K/V LDS reused next tile
+        __syncthreads();
     }
 
-    // ---- epilogue: divide by l, store o[b, qrow, h, d] ----
+    // ---- epilogue: O[q][d] = O^T / l ----
+    if (qvalid) {
+        const float inv = l_run > 0.f ? 1.f / l_run : 0.f;
+        uint16_t* op = p.o + ((int64_t)b * p.Lq + qrow) * (p.H * D) + (int64_t)h * D;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int64_t row = q0 + wave * QB + hi * 4 + r;
-        if (row >= p.Lq) continue;
-        const float inv = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+        for (int dt = 0; dt < 2; ++dt)
 #pragma unroll
-        for (int dblk = 0; dblk < 4; ++dblk) {
-            const int64_t off =
-                ((int64_t)b * p.Lq + row) * (p.H * D) + (int64_t)h * D + dblk * 16 + lo;
-            p.o[off] = __builtin_bit_cast(uint16_t, __float2bfloat16(oacc[dblk][r] * inv));
-        }
+            for (int r = 0; r < 16; ++r) {
+                const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                op[d] = __builtin_bit_cast(uint16_t, __float2bfloat16(ot[dt][r] * inv));
+            }
     }
 }
 
@@ -222,7 +231,7 @@ void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
    hipLaunchKernelGGL(( flash_attn_d64_kernel), dim3(grid), dim3(block), 0, stream, p);
 }
 
-// ---- fragment-layout probe (tests/test_ops_gpu.py verifies the A/B maps) ---
+// ---- fragment-layout probes (tests/test_ops_gpu.py) ------------------------
 namespace {
 __global__ void mfma_probe_kernel(const float* __restrict__ a, const float* __restrict__ b,
                                   float* __restrict__ d) {
@@ -233,13 +242,32 @@ __global__ void mfma_probe_kernel(const float* __restrict__ a, const float* __re
         af[j] = __builtin_bit_cast(short, __float2bfloat16(a[lane * 8 + j]));
         bf[j] = __builtin_bit_cast(short, __float2bfloat16(b[lane * 8 + j]));
     }
-    float4v_ acc = {0.f, 0.f, 0.f, 0.f};
+    float4v_ acc = {};
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
 #pragma unroll
     for (int r = 0; r < 4; ++r) d[lane * 4 + r] = acc[r];
+}
+
+__global__ void mfma_probe32_kernel(const float* __restrict__ a, const float* __restrict__ b,
+                                    float* __restrict__ d) {
+    const int lane = threadIdx.x;
+    short8 af, bf;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        af[j] = __builtin_bit_cast(short, __float2bfloat16(a[lane * 8 + j]));
+        bf[j] = __builtin_bit_cast(short, __float2bfloat16(b[lane * 8 + j]));
+    }
+    float16v acc = {};
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) d[lane * 16 + r] = acc[r];
 }
 }  // namespace
 
 void launch_mfma_probe(const float* a, const float* b, float* d, hipStream_t stream) {
    hipLaunchKernelGGL(( mfma_probe_kernel), dim3(1), dim3(WAVE_SIZE), 0, stream, a, b, d);
+}
+
+void launch_mfma_probe32(const float* a, const float* b, float* d, hipStream_t stream) {
+   hipLaunchKernelGGL(( mfma_probe32_kernel), dim3(1), dim3(WAVE_SIZE), 0, stream, a, b, d);
 }

@@ -170,6 +170,7 @@ std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b);
 
 // defined in attention.hip
 void launch_mfma_probe(const float* a, const float* b, float* d, hipStream_t stream);
+void launch_mfma_probe32(const float* a, const float* b, float* d, hipStream_t stream);
 
 namespace {
 std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b) {
@@ -178,6 +179,15 @@ std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b) {
     auto d = at::zeros({64, 4}, af.options());
     launch_mfma_probe(af.data_ptr<float>(), bf.data_ptr<float>(), d.data_ptr<float>(),
                       cur_stream());
+    return {d};
+}
+
+std::vector<at::Tensor> mfma_probe32(const at::Tensor& a, const at::Tensor& b) {
+    TORCH_CHECK(a.is_cuda() && a.sizes() == (at::IntArrayRef{64, 8}));
+    auto af = a.to(at::kFloat).contiguous(), bf = b.to(at::kFloat).contiguous();
+    auto d = at::zeros({64, 16}, af.options());
+    launch_mfma_probe32(af.data_ptr<float>(), bf.data_ptr<float>(), d.data_ptr<float>(),
+                        cur_stream());
     return {d};
 }
 }  // namespace
@@ -189,4 +199,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("geglu", &geglu, "a * gelu(gate) over last-dim halves");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
     m.def("mfma_probe", &mfma_probe, "dump mfma_f32_16x16x32_bf16 fragment mapping");
+    m.def("mfma_probe32", &mfma_probe32, "dump mfma_f32_32x32x16_bf16 fragment mapping");
 }

@@ -56,6 +56,34 @@ def test_mfma_fragment_layout():
             )
 
 
+@requires_gpu
+def test_mfma32_fragment_layout():
+    """mfma_f32_32x32x16_bf16: A[row=l&31][k=(l>>5)*8+j],
+    B[k=(l>>5)*8+j][col=l&31], D[row=(r&3)+8*(r>>2)+4*(l>>5)][col=l&31]
+    (C/D map verified in the CDNA4 guide §3; A/B checked here)."""
+    torch.manual_seed(1)
+    a_frag = torch.randint(-8, 8, (64, 8), dtype=torch.float32, device=_dev())
+    b_frag = torch.randint(-8, 8, (64, 8), dtype=torch.float32, device=_dev())
+    (d,) = hip_ext().mfma_probe32(a_frag, b_frag)
+    d = d.cpu()
+
+    A = torch.zeros(32, 16)
+    B = torch.zeros(16, 32)
+    for lane in range(64):
+        lo, hi = lane & 31, lane >> 5
+        for j in range(8):
+            A[lo, hi * 8 + j] = a_frag[lane, j]
+            B[hi * 8 + j, lo] = b_frag[lane, j]
+    D = A @ B
+    for lane in range(64):
+        lo, hi = lane & 31, lane >> 5
+        for r in range(16):
+            row = (r & 3) + 8 * (r >> 2) + 4 * hi
+            assert abs(d[lane, r].item() - D[row, lo].item()) < 1e-3, (
+                f"lane {lane} r {r}: got {d[lane, r]}, want {D[row, lo]}"
+            )
+
+
 def _attn_ref(q, k, v):
     return eager.flash_attention(q.float(), k.float(), v.float())
 
