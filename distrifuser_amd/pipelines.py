@@ -87,21 +87,24 @@ class _DistriPipelineBase:
     def _denoise(self, latents, prompt_embeds, added_cond_kwargs, num_inference_steps, guidance_scale):
         cfg = self.distri_config
         do_cfg = cfg.do_classifier_free_guidance
+        from .utils.tracing import trace_range
+
         self.scheduler.set_timesteps(num_inference_steps, device=None)
         self.unet.set_counter(0)
-        for t in _maybe_tqdm(self.scheduler.timesteps, self._progress):
-            latent_in = torch.cat([latents] * 2) if do_cfg else latents
-            latent_in = self.scheduler.scale_model_input(latent_in, t)
-            noise = self.unet(
-                latent_in,
-                t.to(cfg.device) if torch.is_tensor(t) else t,
-                prompt_embeds,
-                added_cond_kwargs,
-            )
-            if do_cfg:
-                n_uncond, n_cond = noise.chunk(2)
-                noise = n_uncond + guidance_scale * (n_cond - n_uncond)
-            latents = self.scheduler.step(noise, t, latents)
+        for i, t in enumerate(_maybe_tqdm(self.scheduler.timesteps, self._progress)):
+            with trace_range(f"denoise_step_{i}"):
+                latent_in = torch.cat([latents] * 2) if do_cfg else latents
+                latent_in = self.scheduler.scale_model_input(latent_in, t)
+                noise = self.unet(
+                    latent_in,
+                    t.to(cfg.device) if torch.is_tensor(t) else t,
+                    prompt_embeds,
+                    added_cond_kwargs,
+                )
+                if do_cfg:
+                    n_uncond, n_cond = noise.chunk(2)
+                    noise = n_uncond + guidance_scale * (n_cond - n_uncond)
+                latents = self.scheduler.step(noise, t, latents)
         return latents
 
     def _decode(self, latents: torch.Tensor, output_type: str):
@@ -164,8 +167,12 @@ class _DistriPipelineBase:
         return [0]
 
     def _capture_graphs(self, static):
+        import torch.distributed as dist
+
         unet = self.unet
+        cfg = self.distri_config
         graphs, outputs = [], []
+        ok = True
         try:
             torch.cuda.synchronize()
             for counter in self._graph_counters():
@@ -175,12 +182,20 @@ class _DistriPipelineBase:
                     out = unet(**static, record=True)
                 graphs.append(g)
                 outputs.append(out)
-            unet.setup_cuda_graph(outputs, graphs)
         except Exception as exc:  # pragma: no cover - depends on RCCL graph support
-            if self.distri_config.rank == 0:
+            ok = False
+            if cfg.rank == 0:
                 print(f"[distrifuser_amd] hipGraph capture failed ({exc}); running eager")
+        # Consensus: if ANY rank failed to capture, all ranks must run eager,
+        # otherwise collective counts diverge and the job deadlocks.
+        if cfg.world_size > 1 and dist.is_initialized():
+            flag = torch.tensor([0 if ok else 1], device=cfg.device)
+            dist.all_reduce(flag, op=dist.ReduceOp.MAX)
+            ok = flag.item() == 0
+        if ok:
+            unet.setup_cuda_graph(outputs, graphs)
+        else:
             unet.setup_cuda_graph(None, None)
-            unet.cuda_graphs = None
             if self.comm_manager is not None:
                 self.comm_manager.clear()
 

@@ -19,10 +19,16 @@ single batched async all-gather — but is laid out for MI355X:
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.distributed as dist
 
 from .config import DistriConfig
+
+# Race-debug mode: make every stale-activation gather synchronous
+# so any handle-lifecycle bug shows up as a numerics diff, not a race.
+DEBUG_SYNC = os.environ.get("DFA_DEBUG_SYNC", "0") == "1"
 
 
 class PatchParallelismCommManager:
@@ -42,6 +48,9 @@ class PatchParallelismCommManager:
 
         # batching state
         self.idx_queue: list[int] = []
+
+        # observability: per-collective accounting (bytes pushed per rank)
+        self.stats = {"gathers": 0, "bytes": 0}
 
     # -- registration pass -------------------------------------------------
 
@@ -143,10 +152,13 @@ class PatchParallelismCommManager:
         own = self.distri_config.split_idx()
         tensor_list = [self.buffer[p, start:end] for p in range(self.buffer.shape[0])]
         handle = dist.all_gather(
-            tensor_list, self.buffer[own, start:end], group=group, async_op=True
+            tensor_list, self.buffer[own, start:end], group=group,
+            async_op=not DEBUG_SYNC,
         )
+        self.stats["gathers"] += 1
+        self.stats["bytes"] += (end - start) * self.buffer.element_size()
         for idx in self.idx_queue:
-            self.handles[idx] = handle
+            self.handles[idx] = handle if not DEBUG_SYNC else None
         self.idx_queue = []
 
     def wait(self, idx: int) -> None:
