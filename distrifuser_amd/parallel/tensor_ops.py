@@ -93,10 +93,12 @@ class TPAttention(nn.Module):
         b, l, _ = x.shape
         ctx = x if encoder_hidden_states is None else encoder_hidden_states
         if self.h_count:
+            from .. import ops
+
             q = self.to_q(x).view(b, l, self.h_count, self.dim_head).transpose(1, 2)
             k = self.to_k(ctx).view(b, ctx.shape[1], self.h_count, self.dim_head).transpose(1, 2)
             v = self.to_v(ctx).view(b, ctx.shape[1], self.h_count, self.dim_head).transpose(1, 2)
-            out = F.scaled_dot_product_attention(q, k, v)
+            out = ops.flash_attention(q, k, v)
             out = out.transpose(1, 2).reshape(b, l, self.h_count * self.dim_head)
             out = F.linear(out, self.to_out_weight)  # bias deferred past the reduce
         else:
