@@ -49,121 +49,15 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
     return r;
 }
 
-// Online softmax runs in exp2 domain: the MFMA scores are scaled by
-// scale*log2(e) once, so p = exp2(s2 - m2) is a single v_exp_f32 per
-// element (no per-element ln2 multiply). Defer-max (guide T13): the O/l
-// rescale is skipped while the tile max stays within DEFER_THR of the
-// running max (p values then bounded by 2^DEFER_THR, safe in f32 accum).
-constexpr float DEFER_THR = 11.0f;  // ~= 8 nats in log2 units
-
-struct WaveState {
-    float m2;     // running max (log2 domain)
-    float l_run;  // running sum
-    float16v ot[2];
-};
-
-template <bool TAIL>
-__device__ __forceinline__ void attn_subtile(
-    const char* k_lds, const char* vt_lds, const short8 (&qf)[4], WaveState& st_state,
-    int st, int lo, int hi, float scale2, int64_t t0, int64_t Lkv) {
-    float16v s = {};
-#pragma unroll
-    for (int ks = 0; ks < 4; ++ks) {
-        const int t = st * 32 + lo;
-        short8 kfrag = *reinterpret_cast<const short8*>(
-            &k_lds[t * 128 + swz(t, (ks * 16 + hi * 8) * 2)]);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
-    }
-    // lane holds S^T rows crow(r) = (r&3)+8*(r>>2)+4*hi for q column lo
-    float tm = -1e30f;
-    float pv[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-        float v = s[r] * scale2;
-        if (TAIL) {
-            const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-            if (t0 + st * 32 + crow >= Lkv) v = -1e30f;
-        }
-        pv[r] = v;
-        tm = fmaxf(tm, v);
-    }
-    tm = fmaxf(tm, __shfl_xor(tm, 32, 64));  // partner holds the other 16 rows
-
-    float& m2 = st_state.m2;
-    // wave-uniform branch (guide T13 recipe): rescale only when some lane's
-    // tile max outgrew its running max by more than the defer threshold
-    const bool need = (m2 == -1e30f) || (tm > m2 + DEFER_THR);
-    if (__any(need)) {
-        const float m_new = fmaxf(m2, tm);
-        const float corr = __builtin_amdgcn_exp2f(m2 - m_new);
-        m2 = m_new;
-        st_state.l_run *= corr;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-            st_state.ot[0][r] *= corr;
-            st_state.ot[1][r] *= corr;
-        }
-    }
-    float tsum = 0.f;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-        pv[r] = __builtin_amdgcn_exp2f(pv[r] - m2);
-        tsum += pv[r];
-    }
-    tsum += __shfl_xor(tsum, 32, 64);
-    st_state.l_run += tsum;
-
-    // ---- pack P to bf16 B-fragments (guide T12: cvt_pk + partner
-    // exchange); B frag j=0..7 -> P^T rows kt*16 + hi*8 + j ----
-    uint32_t w[8], wp[8];
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-        w[i] = cvt_pk_bf16(pv[2 * i], pv[2 * i + 1]);
-        wp[i] = __shfl_xor(w[i], 32, 64);
-    }
-    short8 pb[2];
-#pragma unroll
-    for (int kt = 0; kt < 2; ++kt) {
-        // own word i holds regs {2i,2i+1} = rows {(2i&3)+8*(2i>>2)+4*hi, +1};
-        // hi=0 needs [own 4kt,4kt+1 | partner 4kt,4kt+1], hi=1 the mirror.
-        uint32_t* pbw = reinterpret_cast<uint32_t*>(&pb[kt]);
-        if (hi == 0) {
-            pbw[0] = w[4 * kt + 0];
-            pbw[1] = w[4 * kt + 1];
-            pbw[2] = wp[4 * kt + 0];
-            pbw[3] = wp[4 * kt + 1];
-        } else {
-            pbw[0] = wp[4 * kt + 2];
-            pbw[1] = wp[4 * kt + 3];
-            pbw[2] = w[4 * kt + 2];
-            pbw[3] = w[4 * kt + 3];
-        }
-    }
-
-    // ---- O^T += V^T x P^T ----
-#pragma unroll
-    for (int dt = 0; dt < 2; ++dt) {
-#pragma unroll
-        for (int kt = 0; kt < 2; ++kt) {
-            const int d = dt * 32 + lo;
-            short8 vf = *reinterpret_cast<const short8*>(
-                &vt_lds[d * 128 + swz(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
-            st_state.ot[dt] =
-                __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], st_state.ot[dt], 0, 0, 0);
-        }
-    }
-}
-
 __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(FlashAttnParams p) {
-    // double-buffered K/V tiles: stage tile i+1 while computing tile i
-    __shared__ char k_lds[2][KVB * D * 2];   // [t][d] bf16, swizzled rows
-    __shared__ char vt_lds[2][D * KVB * 2];  // [d][t] bf16, swizzled rows
+    __shared__ char k_lds[KVB * D * 2];   // [t][d] bf16, swizzled rows
+    __shared__ char vt_lds[D * KVB * 2];  // [d][t] bf16, swizzled rows
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
     const int lane = tid % WAVE_SIZE;
-    const int lo = lane & 31;
-    const int hi = lane >> 5;
+    const int lo = lane & 31;   // q column of this lane
+    const int hi = lane >> 5;   // partner-half index
 
     const int bh = blockIdx.y;
     const int b = bh / p.H;
@@ -175,6 +69,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
     const uint16_t* kbase = p.k + b * p.k_sb + h * p.k_sh;
     const uint16_t* vbase = p.v + b * p.v_sb + h * p.v_sh;
 
+    // ---- Q fragments: qf[ks] = Q[q = q0 + wave*32 + lo][d = ks*16 + hi*8 ..] ----
     short8 qf[4];
     const int64_t qrow = q0 + wave * QW + lo;
     const bool qvalid = qrow < p.Lq;
@@ -185,83 +80,147 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
             qf[ks] = *reinterpret_cast<const short8*>(qp + ks * 16 + hi * 8);
     }
 
-    WaveState st_state;
-    st_state.m2 = -1e30f;
-    st_state.l_run = 0.f;
-    st_state.ot[0] = {};
-    st_state.ot[1] = {};
-    const float scale2 = p.scale * 1.44269504088896340736f;  // fold log2(e)
-
-    const int stage_t = tid / 8;  // staging row pair
-    const int stage_d8 = tid % 8;
-
-    auto load_tile = [&](int64_t t0, uint4 (&kraw)[2], uint4 (&vraw)[2]) {
-#pragma unroll
-        for (int rep = 0; rep < 2; ++rep) {
-            const int64_t t_glob = t0 + stage_t + rep * 32;
-            kraw[rep] = {0, 0, 0, 0};
-            vraw[rep] = {0, 0, 0, 0};
-            if (t_glob < Lkv) {
-                const int64_t chunk = t_glob / p.LC;
-                const int64_t tin = t_glob % p.LC;
-                kraw[rep] = *reinterpret_cast<const uint4*>(
-                    kbase + chunk * p.k_sc + tin * p.k_sl + stage_d8 * 8);
-                vraw[rep] = *reinterpret_cast<const uint4*>(
-                    vbase + chunk * p.v_sc + tin * p.v_sl + stage_d8 * 8);
-            }
-        }
-    };
-    auto write_tile = [&](int buf, const uint4 (&kraw)[2], const uint4 (&vraw)[2]) {
-#pragma unroll
-        for (int rep = 0; rep < 2; ++rep) {
-            const int t_local = stage_t + rep * 32;
-            *reinterpret_cast<uint4*>(
-                &k_lds[buf][t_local * 128 + swz(t_local, stage_d8 * 16)]) = kraw[rep];
-            const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vraw[rep]);
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int d = stage_d8 * 8 + j;
-                *reinterpret_cast<uint16_t*>(
-                    &vt_lds[buf][d * 128 + swz(d, t_local * 2)]) = ve[j];
-            }
-        }
-    };
+    float m_run = -1e30f;
+    float l_run = 0.f;
+    float16v ot[2] = {};  // O^T tiles: [dt] -> rows d = dt*32 + crow(r,hi), col q=lo
+    // exp2-domain softmax: v_exp_f32 IS exp2 — fold log2(e) into the scale
+    // (saves one v_mul per score element over __expf's lowering)
+    const float scale2 = p.scale * 1.44269504088896340736f;
 
     const int n_tiles = (int)((Lkv + KVB - 1) / KVB);
-    const int n_full = (int)(Lkv / KVB);  // tiles with no tail masking
-
-    uint4 kraw[2], vraw[2];
-    load_tile(0, kraw, vraw);
-    write_tile(0, kraw, vraw);
-    __syncthreads();
-
     for (int tile = 0; tile < n_tiles; ++tile) {
         const int64_t t0 = (int64_t)tile * KVB;
-        const int buf = tile & 1;
-        if (tile + 1 < n_tiles)  // issue next tile's global loads early
-            load_tile(t0 + KVB, kraw, vraw);
-        if (tile < n_full) {
-            attn_subtile<false>(k_lds[buf], vt_lds[buf], qf, st_state, 0, lo, hi, scale2, t0, Lkv);
-            attn_subtile<false>(k_lds[buf], vt_lds[buf], qf, st_state, 1, lo, hi, scale2, t0, Lkv);
-        } else {
-            attn_subtile<true>(k_lds[buf], vt_lds[buf], qf, st_state, 0, lo, hi, scale2, t0, Lkv);
-            attn_subtile<true>(k_lds[buf], vt_lds[buf], qf, st_state, 1, lo, hi, scale2, t0, Lkv);
+        // ---- cooperative staging: K [t][d], V^T [d][t] ----
+        {
+            const int tl = tid / 8;
+            const int d8 = tid % 8;
+#pragma unroll
+            for (int rep = 0; rep < 2; ++rep) {
+                const int t_local = tl + rep * 32;
+                const int64_t t_glob = t0 + t_local;
+                uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
+                if (t_glob < Lkv) {
+                    const int64_t chunk = t_glob / p.LC;
+                    const int64_t tin = t_glob % p.LC;
+                    kraw = *reinterpret_cast<const uint4*>(
+                        kbase + chunk * p.k_sc + tin * p.k_sl + d8 * 8);
+                    vraw = *reinterpret_cast<const uint4*>(
+                        vbase + chunk * p.v_sc + tin * p.v_sl + d8 * 8);
+                }
+                *reinterpret_cast<uint4*>(&k_lds[t_local * 128 + swz(t_local, d8 * 16)]) = kraw;
+                const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vraw);
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int d = d8 * 8 + j;
+                    *reinterpret_cast<uint16_t*>(&vt_lds[d * 128 + swz(d, t_local * 2)]) = ve[j];
+                }
+            }
         }
-        if (tile + 1 < n_tiles)
-            write_tile(buf ^ 1, kraw, vraw);
+        __syncthreads();
+
+#pragma unroll
+        for (int st = 0; st < 2; ++st) {  // two 32-token sub-tiles
+            // ---- S^T[kv32][q32] = K_sub x Q^T ----
+            float16v s = {};
+#pragma unroll
+            for (int ks = 0; ks < 4; ++ks) {
+                const int t = st * 32 + lo;
+                short8 kfrag = *reinterpret_cast<const short8*>(
+                    &k_lds[t * 128 + swz(t, (ks * 16 + hi * 8) * 2)]);
+                s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
+            }
+            // lane holds S^T rows crow(r) = (r&3)+8*(r>>2)+4*hi for q col lo
+            float tm = -1e30f;
+            float pv[16];
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const int64_t tglob = t0 + st * 32 + crow;
+                float v = (tglob < Lkv) ? s[r] * scale2 : -1e30f;
+                pv[r] = v;
+                tm = fmaxf(tm, v);
+            }
+            tm = fmaxf(tm, __shfl_xor(tm, 32, 64));  // partner holds the other 16 rows
+            const float m_new = fmaxf(m_run, tm);
+            const float corr = __builtin_amdgcn_exp2f(m_run - m_new);
+            m_run = m_new;
+            l_run *= corr;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                ot[0][r] *= corr;
+                ot[1][r] *= corr;
+            }
+            float tsum = 0.f;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                pv[r] = __builtin_amdgcn_exp2f(pv[r] - m_new);
+                tsum += pv[r];
+            }
+            tsum += __shfl_xor(tsum, 32, 64);
+            l_run += tsum;
+
+            // ---- pack P to bf16 B-fragments (guide T12: cvt_pk + partner
+            // exchange; P^T[t][q] needs rows hi*8..hi*8+7 per k-slice) ----
+            uint32_t w[8], wp[8];
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                w[i] = cvt_pk_bf16(pv[2 * i], pv[2 * i + 1]);
+                wp[i] = __shfl_xor(w[i], 32, 64);
+            }
+            short8 pb[2];
+#pragma unroll
+            for (int kt = 0; kt < 2; ++kt) {
+                // B frag j=0..7 -> P^T rows kt*16 + hi*8 + j.
+                // own regs r cover rows (r&3)+8*(r>>2)+4*hi; packed word i
+                // holds regs {2i, 2i+1} = rows {(2i&3)+8*(2i>>2)+4*hi, ...}.
+                // For hi=0: rows kt*16+0..7 = own words {4kt,4kt+1} (rows
+                // kt*16+0..3) then partner words {4kt,4kt+1} (rows kt*16+4..7).
+                // For hi=1: rows kt*16+8..15 = partner {4kt+2,4kt+3} then own
+                // {4kt+2,4kt+3}.
+                uint32_t b0, b1, b2, b3;
+                if (hi == 0) {
+                    b0 = w[4 * kt + 0];
+                    b1 = w[4 * kt + 1];
+                    b2 = wp[4 * kt + 0];
+                    b3 = wp[4 * kt + 1];
+                } else {
+                    b0 = wp[4 * kt + 2];
+                    b1 = wp[4 * kt + 3];
+                    b2 = w[4 * kt + 2];
+                    b3 = w[4 * kt + 3];
+                }
+                uint32_t* pbw = reinterpret_cast<uint32_t*>(&pb[kt]);
+                pbw[0] = b0;
+                pbw[1] = b1;
+                pbw[2] = b2;
+                pbw[3] = b3;
+            }
+
+            // ---- O^T += V^T x P^T ----
+#pragma unroll
+            for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+                for (int kt = 0; kt < 2; ++kt) {
+                    const int d = dt * 32 + lo;
+                    short8 vf = *reinterpret_cast<const short8*>(
+                        &vt_lds[d * 128 + swz(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
+                    ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
+                }
+            }
+        }
         __syncthreads();
     }
 
+    // ---- epilogue: O[q][d] = O^T / l ----
     if (qvalid) {
-        const float inv = st_state.l_run > 0.f ? 1.f / st_state.l_run : 0.f;
+        const float inv = l_run > 0.f ? 1.f / l_run : 0.f;
         uint16_t* op = p.o + ((int64_t)b * p.Lq + qrow) * (p.H * D) + (int64_t)h * D;
 #pragma unroll
         for (int dt = 0; dt < 2; ++dt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-                op[d] = __builtin_bit_cast(uint16_t,
-                                           __float2bfloat16(st_state.ot[dt][r] * inv));
+                op[d] = __builtin_bit_cast(uint16_t, __float2bfloat16(ot[dt][r] * inv));
             }
     }
 }

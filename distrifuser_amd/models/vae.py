@@ -60,9 +60,14 @@ class VAEResnetBlock(nn.Module):
 
 
 def _chunked_single_head_attention(
-    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, chunk: int = 16384
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, chunk: int = 4096
 ) -> torch.Tensor:
-    """[B, L, C] single-head attention with O(chunk*L) memory."""
+    """[B, L, C] single-head attention with O(chunk*L) memory.
+
+    At 3840^2 the VAE mid block attends over 480*480 = 230k tokens with a
+    single 512-dim head; the score slab per chunk is chunk*L fp32
+    (4096*230k*4B ~= 3.8 GB) — bounded regardless of resolution.
+    """
     scale = q.shape[-1] ** -0.5
     outs = []
     for s in range(0, q.shape[1], chunk):
