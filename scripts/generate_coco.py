@@ -5,6 +5,12 @@ environment has no network — `--caption_file` takes a JSON list of strings or
 a prompts.json produced by scripts/dump_coco.py; without it a deterministic
 synthetic caption set is used)."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import json
 import os

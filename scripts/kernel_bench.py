@@ -3,6 +3,12 @@
 Run on an MI355X: python scripts/kernel_bench.py > gpurun_out/kernel_bench.json
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import json
 import sys
 import time

@@ -2,6 +2,12 @@
 profile_macs.py, natively — forward hooks over our own modules instead of
 torchprofile)."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 
 import torch

@@ -4,6 +4,12 @@ scripts/run_sdxl.py flag surface; launched under torchrun for N>1):
   torchrun --nproc_per_node=N scripts/run_sdxl.py --mode benchmark ...
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import time
 

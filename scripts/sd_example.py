@@ -1,5 +1,11 @@
 """Minimal SD1.5 usage example (parity with the reference's sd_example.py)."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import torch
 
 from distrifuser_amd import DistriConfig, DistriSDPipeline

@@ -3,6 +3,12 @@
   torchrun --nproc_per_node=2 scripts/sdxl_example.py
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import torch
 
 from distrifuser_amd import DistriConfig, DistriSDXLPipeline
