@@ -34,7 +34,7 @@ namespace {
 constexpr int QW = 32;     // q rows per wave
 constexpr int WAVES = 4;   // waves per block
 constexpr int QBLK = QW * WAVES;  // 128
-constexpr int KVB = 64;    // kv tokens per LDS tile
+constexpr int KVB = 128;   // kv tokens per LDS tile
 constexpr int D = 64;
 
 typedef float float4v_ __attribute__((ext_vector_type(4)));
@@ -50,6 +50,7 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
     return r;
 }
 
+template <bool MASK, bool DEFER>
 __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(FlashAttnParams p) {
     __shared__ char k_lds[KVB * D * 2];   // [t][d] bf16, swizzled rows
     __shared__ char vt_lds[D * KVB * 2];  // [d][t] bf16, swizzled rows
@@ -81,12 +82,19 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
             qf[ks] = *reinterpret_cast<const short8*>(qp + ks * 16 + hi * 8);
     }
 
-    float m_run = -1e30f;
+    // Online softmax bookkeeping (exp2 domain, guide §B):
+    //   m_raw   running max of RAW scores (max commutes with the positive
+    //           scale, so per-element work is p = exp2(fma(s, scale2, -msc))
+    //           — ONE v_fma + ONE v_exp per score element)
+    //   msc     m_raw * scale2 (refreshed only when m_raw moves)
+    //   defer-max (guide T13): skip the O/l rescale while the tile max stays
+    //   within DEFER_THR (log2 units) of the running max
+    float m_raw = -1e30f;
+    float msc = -1e30f;
     float l_run = 0.f;
     float16v ot[2] = {};  // O^T tiles: [dt] -> rows d = dt*32 + crow(r,hi), col q=lo
-    // exp2-domain softmax: v_exp_f32 IS exp2 — fold log2(e) into the scale
-    // (saves one v_mul per score element over __expf's lowering)
     const float scale2 = p.scale * 1.44269504088896340736f;
+    const float defer_raw = 11.0f / scale2;
 
     const int n_tiles = (int)((Lkv + KVB - 1) / KVB);
     for (int tile = 0; tile < n_tiles; ++tile) {
@@ -96,11 +104,11 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
             const int tl = tid / 8;
             const int d8 = tid % 8;
 #pragma unroll
-            for (int rep = 0; rep < 2; ++rep) {
+            for (int rep = 0; rep < KVB / 32; ++rep) {
                 const int t_local = tl + rep * 32;
                 const int64_t t_glob = t0 + t_local;
                 uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
-                if (t_glob < Lkv) {
+                if (!MASK || t_glob < Lkv) {
                     const int64_t chunk = t_glob / p.LC;
                     const int64_t tin = t_glob % p.LC;
                     kraw = *reinterpret_cast<const uint4*>(
@@ -120,9 +128,10 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
         __syncthreads();
 
 #pragma unroll
-        for (int st = 0; st < 2; ++st) {  // two 32-token sub-tiles
+        for (int st = 0; st < KVB / 32; ++st) {  // 32-token sub-tiles
             // ---- S^T[kv32][q32] = K_sub x Q^T ----
             float16v s = {};
+            __builtin_amdgcn_s_setprio(1);  // favor the MFMA wave (guide T5)
 #pragma unroll
             for (int ks = 0; ks < 4; ++ks) {
                 const int t = st * 32 + lo;
@@ -130,38 +139,46 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
                     &k_lds[t * 128 + swz(t, (ks * 16 + hi * 8) * 2)]);
                 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
             }
+            __builtin_amdgcn_s_setprio(0);
             // lane holds S^T rows crow(r) = (r&3)+8*(r>>2)+4*hi for q col lo
             float tm = -1e30f;
             float pv[16];
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-                const int64_t tglob = t0 + st * 32 + crow;
-                float v = (tglob < Lkv) ? s[r] * scale2 : -1e30f;
+                float v = s[r];
+                if (MASK) {
+                    const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+                    if (t0 + st * 32 + crow >= Lkv) v = -1e30f;
+                }
                 pv[r] = v;
                 tm = fmaxf(tm, v);
             }
             tm = fmaxf(tm, __shfl_xor(tm, 32, 64));  // partner holds the other 16 rows
-            const float m_new = fmaxf(m_run, tm);
-            const float corr = __builtin_amdgcn_exp2f(m_run - m_new);
-            m_run = m_new;
-            l_run *= corr;
+
+            if (!DEFER || __any(m_raw == -1e30f || tm > m_raw + defer_raw)) {
+                const float m_new = fmaxf(m_raw, tm);
+                const float msc_new = m_new * scale2;
+                const float corr = __builtin_amdgcn_exp2f(msc - msc_new);
+                m_raw = m_new;
+                msc = msc_new;
+                l_run *= corr;
 #pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                ot[0][r] *= corr;
-                ot[1][r] *= corr;
+                for (int r = 0; r < 16; ++r) {
+                    ot[0][r] *= corr;
+                    ot[1][r] *= corr;
+                }
             }
             float tsum = 0.f;
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                pv[r] = __builtin_amdgcn_exp2f(pv[r] - m_new);
+                pv[r] = __builtin_amdgcn_exp2f(__builtin_fmaf(pv[r], scale2, -msc));
                 tsum += pv[r];
             }
             tsum += __shfl_xor(tsum, 32, 64);
             l_run += tsum;
 
             // ---- pack P to bf16 B-fragments (guide T12: cvt_pk + partner
-            // exchange; P^T[t][q] needs rows hi*8..hi*8+7 per k-slice) ----
+            // exchange); B frag j=0..7 -> P^T rows kt*16 + hi*8 + j ----
             uint32_t w[8], wp[8];
 #pragma unroll
             for (int i = 0; i < 8; ++i) {
@@ -171,33 +188,24 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
             short8 pb[2];
 #pragma unroll
             for (int kt = 0; kt < 2; ++kt) {
-                // B frag j=0..7 -> P^T rows kt*16 + hi*8 + j.
-                // own regs r cover rows (r&3)+8*(r>>2)+4*hi; packed word i
-                // holds regs {2i, 2i+1} = rows {(2i&3)+8*(2i>>2)+4*hi, ...}.
-                // For hi=0: rows kt*16+0..7 = own words {4kt,4kt+1} (rows
-                // kt*16+0..3) then partner words {4kt,4kt+1} (rows kt*16+4..7).
-                // For hi=1: rows kt*16+8..15 = partner {4kt+2,4kt+3} then own
-                // {4kt+2,4kt+3}.
-                uint32_t b0, b1, b2, b3;
-                if (hi == 0) {
-                    b0 = w[4 * kt + 0];
-                    b1 = w[4 * kt + 1];
-                    b2 = wp[4 * kt + 0];
-                    b3 = wp[4 * kt + 1];
-                } else {
-                    b0 = wp[4 * kt + 2];
-                    b1 = wp[4 * kt + 3];
-                    b2 = w[4 * kt + 2];
-                    b3 = w[4 * kt + 3];
-                }
+                // own word i holds regs {2i,2i+1} = rows {(2i&3)+8*(2i>>2)+4*hi, +1};
+                // hi=0 needs [own 4kt,4kt+1 | partner 4kt,4kt+1], hi=1 the mirror.
                 uint32_t* pbw = reinterpret_cast<uint32_t*>(&pb[kt]);
-                pbw[0] = b0;
-                pbw[1] = b1;
-                pbw[2] = b2;
-                pbw[3] = b3;
+                if (hi == 0) {
+                    pbw[0] = w[4 * kt + 0];
+                    pbw[1] = w[4 * kt + 1];
+                    pbw[2] = wp[4 * kt + 0];
+                    pbw[3] = wp[4 * kt + 1];
+                } else {
+                    pbw[0] = wp[4 * kt + 2];
+                    pbw[1] = wp[4 * kt + 3];
+                    pbw[2] = w[4 * kt + 2];
+                    pbw[3] = w[4 * kt + 3];
+                }
             }
 
             // ---- O^T += V^T x P^T ----
+            __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int dt = 0; dt < 2; ++dt) {
 #pragma unroll
@@ -208,6 +216,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
                     ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
                 }
             }
+            __builtin_amdgcn_s_setprio(0);
         }
         __syncthreads();
     }
@@ -228,10 +237,24 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
 
 }  // namespace
 
+#include <cstdlib>
+
 void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
     dim3 grid((unsigned)((p.Lq + QBLK - 1) / QBLK), (unsigned)(p.B * p.H));
     dim3 block(WAVES * WAVE_SIZE);
-   hipLaunchKernelGGL(( flash_attn_d64_kernel), dim3(grid), dim3(block), 0, stream, p);
+    static const bool defer = [] {
+        const char* e = std::getenv("DFA_ATTN_DEFER");
+        return e == nullptr || e[0] != '0';  // defer-max on by default
+    }();
+    const bool mask = (p.NC * p.LC) % KVB != 0;
+    if (!mask && defer)
+       hipLaunchKernelGGL(( flash_attn_d64_kernel<false, true>), dim3(grid), dim3(block), 0, stream, p);
+    else if (!mask)
+       hipLaunchKernelGGL(( flash_attn_d64_kernel<false, false>), dim3(grid), dim3(block), 0, stream, p);
+    else if (defer)
+       hipLaunchKernelGGL(( flash_attn_d64_kernel<true, true>), dim3(grid), dim3(block), 0, stream, p);
+    else
+       hipLaunchKernelGGL(( flash_attn_d64_kernel<true, false>), dim3(grid), dim3(block), 0, stream, p);
 }
 
 // ---- fragment-layout probes (tests/test_ops_gpu.py) ------------------------
