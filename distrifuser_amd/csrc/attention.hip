@@ -35,6 +35,8 @@ constexpr int WAVES = 4;   // waves per block
 constexpr int QBLK = QW * WAVES;  // 128
 constexpr int KVB = 128;   // kv tokens per LDS tile
 constexpr int D = 64;
+constexpr int K_ROW = D * 2;    // K LDS row bytes   [t][d]
+constexpr int VT_ROW = KVB * 2; // V^T LDS row bytes [d][t]
 
 typedef float float4v_ __attribute__((ext_vector_type(4)));
 typedef float float16v __attribute__((ext_vector_type(16)));
@@ -115,12 +117,12 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
                     vraw = *reinterpret_cast<const uint4*>(
                         vbase + chunk * p.v_sc + tin * p.v_sl + d8 * 8);
                 }
-                *reinterpret_cast<uint4*>(&k_lds[t_local * 128 + swz(t_local, d8 * 16)]) = kraw;
+                *reinterpret_cast<uint4*>(&k_lds[t_local * K_ROW + swz(t_local, d8 * 16)]) = kraw;
                 const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vraw);
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     const int d = d8 * 8 + j;
-                    *reinterpret_cast<uint16_t*>(&vt_lds[d * 128 + swz(d, t_local * 2)]) = ve[j];
+                    *reinterpret_cast<uint16_t*>(&vt_lds[d * VT_ROW + swz(d, t_local * 2)]) = ve[j];
                 }
             }
         }
@@ -135,7 +137,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
             for (int ks = 0; ks < 4; ++ks) {
                 const int t = st * 32 + lo;
                 short8 kfrag = *reinterpret_cast<const short8*>(
-                    &k_lds[t * 128 + swz(t, (ks * 16 + hi * 8) * 2)]);
+                    &k_lds[t * K_ROW + swz(t, (ks * 16 + hi * 8) * 2)]);
                 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
@@ -211,7 +213,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
                 for (int kt = 0; kt < 2; ++kt) {
                     const int d = dt * 32 + lo;
                     short8 vf = *reinterpret_cast<const short8*>(
-                        &vt_lds[d * 128 + swz(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
+                        &vt_lds[d * VT_ROW + swz(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
                     ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
                 }
             }
