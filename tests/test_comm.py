@@ -92,3 +92,21 @@ def _checkpoint_flush(rank, world_size):
 def test_comm_checkpoint_flush_ws2():
     out = run_distributed(2, _checkpoint_flush)
     assert all(out.values())
+
+
+def _stats_worker(rank, world_size):
+    cfg = DistriConfig(do_classifier_free_guidance=False, device="cpu")
+    comm = PatchParallelismCommManager(cfg)
+    i0 = comm.register_tensor((8,), torch.float32)
+    comm.create_buffer()
+    comm.enqueue(i0, torch.ones(8))
+    comm.communicate()
+    comm.clear()
+    return dict(comm.stats)
+
+
+def test_comm_stats_accounting():
+    out = run_distributed(2, _stats_worker)
+    for r in (0, 1):
+        assert out[r]["gathers"] == 1
+        assert out[r]["bytes"] == 8 * 4
