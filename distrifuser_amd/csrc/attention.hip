@@ -31,7 +31,7 @@
 namespace {
 
 constexpr int QW = 32;     // q rows per wave
-constexpr int WAVES = 4;   // waves per block
+constexpr int WAVES = 8;   // waves per block (QBLK=256: halves KV re-reads)
 constexpr int QBLK = QW * WAVES;  // 128
 constexpr int KVB = 128;   // kv tokens per LDS tile
 constexpr int D = 64;
@@ -102,11 +102,12 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE) void flash_attn_d64_kernel(Flash
         const int64_t t0 = (int64_t)tile * KVB;
         // ---- cooperative staging: K [t][d], V^T [d][t] ----
         {
-            const int tl = tid / 8;
+            const int tl = tid / 8;                      // covers WAVES*8 rows/pass
             const int d8 = tid % 8;
+            constexpr int ROWS_PER_PASS = WAVES * WAVE_SIZE / 8;
 #pragma unroll
-            for (int rep = 0; rep < KVB / 32; ++rep) {
-                const int t_local = tl + rep * 32;
+            for (int rep = 0; rep < KVB / ROWS_PER_PASS; ++rep) {
+                const int t_local = tl + rep * ROWS_PER_PASS;
                 const int64_t t_glob = t0 + t_local;
                 uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
                 if (!MASK || t_glob < Lkv) {
