@@ -63,6 +63,25 @@ class _DistriPipelineBase:
     def set_progress_bar_config(self, disable: bool = False, **kwargs):
         self._progress = not disable and self.distri_config.rank == 0
 
+    def save_pretrained(self, out_dir: str) -> None:
+        """Write a diffusers-layout safetensors directory loadable by
+        ``from_pretrained(pretrained_model_name_or_path=out_dir)``."""
+        import os
+
+        from safetensors.torch import save_file
+
+        from .models.weights import export_diffusers_state_dict
+
+        components = {"unet": self.unet.unet, "vae": self.vae}
+        if hasattr(self, "text_encoder"):
+            components["text_encoder"] = self.text_encoder
+        if hasattr(self, "text_encoder_2"):
+            components["text_encoder_2"] = self.text_encoder_2
+        for name, model in components.items():
+            os.makedirs(os.path.join(out_dir, name), exist_ok=True)
+            sd = {k: v.contiguous().cpu() for k, v in export_diffusers_state_dict(model).items()}
+            save_file(sd, os.path.join(out_dir, name, "model.safetensors"))
+
     @property
     def device(self):
         return self.distri_config.device

@@ -54,6 +54,28 @@ def test_vae_roundtrip():
         assert torch.equal(va, vb), k
 
 
+def test_pipeline_save_and_reload(tmp_path):
+    """save_pretrained -> from_pretrained(path) reproduces outputs exactly."""
+    from distrifuser_amd import DistriSDXLPipeline
+
+    cfg = DistriConfig(height=128, width=128, use_cuda_graph=False, device="cpu")
+    torch.manual_seed(0)
+    pipe = DistriSDXLPipeline.from_pretrained(cfg, preset="tiny", torch_dtype=torch.float32)
+    pipe.save_pretrained(str(tmp_path))
+    assert (tmp_path / "unet" / "model.safetensors").exists()
+
+    torch.manual_seed(123)  # different init; weights must come from disk
+    pipe2 = DistriSDXLPipeline.from_pretrained(
+        cfg, preset="tiny", torch_dtype=torch.float32,
+        pretrained_model_name_or_path=str(tmp_path),
+    )
+    g = torch.Generator().manual_seed(9)
+    out1 = pipe("same picture", num_inference_steps=2, output_type="latent", generator=g)
+    g = torch.Generator().manual_seed(9)
+    out2 = pipe2("same picture", num_inference_steps=2, output_type="latent", generator=g)
+    assert torch.allclose(out1, out2, atol=1e-6), (out1 - out2).abs().max()
+
+
 def test_clip_roundtrip():
     torch.manual_seed(0)
     src = CLIPTextEncoder(TINY_CLIP)
