@@ -53,6 +53,30 @@ def main():
         })
         print(json.dumps(results[-1]), flush=True)
 
+    # ---- displaced-patch shape: local Q x full stale KV (8 chunks) ----
+    for h, lq, n in [(10, 7200, 8), (20, 3600, 4)]:
+        lkv = lq * n
+        inner = h * 64
+        q = torch.randn(1, h, lq, 64, device=dev, dtype=torch.bfloat16)
+        slot = lq * 2 * inner
+        buf = torch.randn(n, slot + 64, device=dev, dtype=torch.bfloat16)
+        kv = buf[:, :slot].view(n, 1, lq, 2 * inner)
+        k5 = kv[..., :inner].unflatten(-1, (h, 64)).permute(1, 3, 0, 2, 4)
+        v5 = kv[..., inner:].unflatten(-1, (h, 64)).permute(1, 3, 0, 2, 4)
+        ms_ours = timeit(lambda: ops.hip_ext().flash_attention(q, k5, v5))
+        kc = kv.permute(1, 0, 2, 3).reshape(1, lkv, 2 * inner)
+        kcat = kc[..., :inner].view(1, lkv, h, 64).transpose(1, 2).contiguous()
+        vcat = kc[..., inner:].view(1, lkv, h, 64).transpose(1, 2).contiguous()
+        ms_sdpa = timeit(lambda: F.scaled_dot_product_attention(q, kcat, vcat))
+        flops = 4.0 * lq * lkv * 64 * h
+        results.append({
+            "op": "flash_attention_stale_chunked", "heads": h, "Lq": lq, "Lkv": lkv,
+            "n_chunks": n, "ms_ours": ms_ours, "ms_sdpa_precat": ms_sdpa,
+            "tflops_ours": flops / ms_ours / 1e9,
+            "tflops_sdpa_precat": flops / ms_sdpa / 1e9,
+        })
+        print(json.dumps(results[-1]), flush=True)
+
     # ---- fused GN+SiLU at SDXL shapes ----
     for c, hw in [(320, 480), (640, 240), (1280, 120), (320, 128), (640, 64)]:
         x = torch.randn(2, c, hw, hw, device=dev, dtype=torch.bfloat16)
