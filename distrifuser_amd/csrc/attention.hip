@@ -264,12 +264,11 @@ static void launch_geom(const FlashAttnParams& p, hipStream_t stream) {
 }
 
 void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
-    // Geometry by query length: long sequences amortize KV re-reads with
-    // 512-row Q blocks; short ones need the extra blocks for occupancy.
-    if (p.Lq >= 24576)
-        launch_geom<16>(p, stream);
-    else
-        launch_geom<8>(p, stream);
+    // 8-wave (256-row) blocks measured best across the SD-family shapes
+    // (16-wave was neutral at L=57.6k: KV re-reads stop being the bound and
+    // the 16-wave barrier domain eats the gain). The geometry template stays
+    // for future tuning.
+    launch_geom<8>(p, stream);
 }
 
 // ---- fragment-layout probes (tests/test_ops_gpu.py) ------------------------
