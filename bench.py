@@ -45,6 +45,11 @@ def main() -> None:
     from distrifuser_amd import DistriConfig, DistriSDXLPipeline
 
     use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        # let MIOpen search for the best conv algorithm per shape (the
+        # default immediate mode picked an im2col path for some 3840^2 convs
+        # — profiles/rocprof_3840_r01.md)
+        torch.backends.cudnn.benchmark = True
     cfg = DistriConfig(
         height=args.height,
         width=args.width,
