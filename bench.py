@@ -38,6 +38,8 @@ def main() -> None:
     ap.add_argument("--mode", type=str, default="corrected_async_gn")
     ap.add_argument("--parallelism", type=str, default="patch")
     ap.add_argument("--no-cuda-graph", action="store_true")
+    ap.add_argument("--force-cuda-graph", action="store_true",
+                    help="capture hipGraphs also at world_size>1 (RCCL-in-graph)")
     ap.add_argument("--no-split-batch", action="store_true")
     ap.add_argument("--warmup-steps", type=int, default=4, help="sync-comm denoise steps")
     args = ap.parse_args()
@@ -50,6 +52,14 @@ def main() -> None:
         # default immediate mode picked an im2col path for some 3840^2 convs
         # — profiles/rocprof_3840_r01.md)
         torch.backends.cudnn.benchmark = True
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    # hipGraph capture is the verified fast path at world_size 1; at 3840^2
+    # the step is kernel-bound and graphs measured neutral, so multi-rank
+    # runs stay eager by default (RCCL-in-graph capture is opt-in via
+    # --force-cuda-graph until verified on an 8-GPU node).
+    use_graphs = use_cuda and not args.no_cuda_graph and (
+        world == 1 or args.force_cuda_graph
+    )
     cfg = DistriConfig(
         height=args.height,
         width=args.width,
@@ -58,7 +68,7 @@ def main() -> None:
         warmup_steps=args.warmup_steps,
         mode=args.mode,
         parallelism=args.parallelism,
-        use_cuda_graph=use_cuda and not args.no_cuda_graph,
+        use_cuda_graph=use_graphs,
     )
     if cfg.world_size > 1:
         assert cfg.world_size == args.gpus, (

@@ -130,6 +130,28 @@ def test_flash_attention_chunked_matches_cat():
 
 
 @requires_gpu
+def test_flash_attention_chunked_batch2():
+    """CFG-unsplit patch case: batch 2 rides the same chunked buffer."""
+    torch.manual_seed(1)
+    dev = _dev()
+    b, heads, dim_head, l, n = 2, 4, 64, 128, 2
+    inner = heads * dim_head
+    q = torch.randn(b, l, inner, device=dev, dtype=torch.bfloat16)
+    slot = b * l * 2 * inner
+    buf = torch.randn(n, slot + 8, device=dev, dtype=torch.bfloat16)
+    kv_chunks = buf[:, :slot].view(n, b, l, 2 * inner)
+    out = ops.flash_attention_chunked(q, kv_chunks, heads, dim_head).float()
+
+    full = kv_chunks.permute(1, 0, 2, 3).reshape(b, n * l, 2 * inner).float()
+    kf, vf = full.split(inner, dim=-1)
+    qf = q.float().view(b, l, heads, dim_head).transpose(1, 2)
+    kf = kf.view(b, n * l, heads, dim_head).transpose(1, 2)
+    vf = vf.view(b, n * l, heads, dim_head).transpose(1, 2)
+    ref = eager.flash_attention(qf, kf, vf).transpose(1, 2).reshape(b, l, inner)
+    assert (out - ref).abs().max().item() < 0.03
+
+
+@requires_gpu
 def test_group_norm_stats_gpu():
     torch.manual_seed(0)
     x = torch.randn(2, 64, 33, 40, device=_dev(), dtype=torch.bfloat16)
