@@ -42,6 +42,8 @@ def main() -> None:
                     help="capture hipGraphs also at world_size>1 (RCCL-in-graph)")
     ap.add_argument("--no-split-batch", action="store_true")
     ap.add_argument("--warmup-steps", type=int, default=4, help="sync-comm denoise steps")
+    ap.add_argument("--preset", type=str, default="sdxl", choices=["sdxl", "tiny"],
+                    help="tiny = CPU plumbing test of the distributed bench path")
     args = ap.parse_args()
 
     from distrifuser_amd import DistriConfig, DistriSDXLPipeline
@@ -77,7 +79,9 @@ def main() -> None:
     dtype = torch.bfloat16 if use_cuda else torch.float32
 
     torch.manual_seed(0)
-    pipe = DistriSDXLPipeline.from_pretrained(cfg, torch_dtype=dtype, scheduler=args.scheduler)
+    pipe = DistriSDXLPipeline.from_pretrained(
+        cfg, torch_dtype=dtype, scheduler=args.scheduler, preset=args.preset
+    )
 
     # ---- manual denoise loop so we can time exactly K steps ----
     total_steps = args.warmup + args.steps
