@@ -44,7 +44,15 @@ typedef float float4v_ __attribute__((ext_vector_type(4)));
 typedef float float16v __attribute__((ext_vector_type(16)));
 
 __device__ __forceinline__ int swz(int row, int byte_off) {
+    // 8-window XOR within a 128 B row (K tile; the 16 B-aligned b128 floor
+    // for 32-row column reads is 4-way here — see guide §6 G4)
     return byte_off ^ ((row & 7) << 4);
+}
+
+__device__ __forceinline__ int swz16(int row, int byte_off) {
+    // 16-window XOR within a 256 B row (V^T tile): spreads the 32-row
+    // column reads 2-way, which is free on the 32-bank LDS (guide m136)
+    return byte_off ^ ((row & 15) << 4);
 }
 
 __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
@@ -126,7 +134,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     const int d = d8 * 8 + j;
-                    *reinterpret_cast<uint16_t*>(&vt_lds[d * VT_ROW + swz(d, t_local * 2)]) = ve[j];
+                    *reinterpret_cast<uint16_t*>(&vt_lds[d * VT_ROW + swz16(d, t_local * 2)]) = ve[j];
                 }
             }
         }
@@ -217,7 +225,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
                 for (int kt = 0; kt < 2; ++kt) {
                     const int d = dt * 32 + lo;
                     short8 vf = *reinterpret_cast<const short8*>(
-                        &vt_lds[d * VT_ROW + swz(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
+                        &vt_lds[d * VT_ROW + swz16(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
                     ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
                 }
             }
