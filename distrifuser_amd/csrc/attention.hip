@@ -60,7 +60,10 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
     return r;
 }
 
-template <int NW, bool MASK, bool DEFER>
+// ABL: perf-ablation variants (wrong numerics, PMC probing only; selected
+// by DFA_ATTN_ABLATE): 1 = V^T staging as a single linear conflict-free
+// b128 write instead of the 8-lane scalar transpose.
+template <int NW, bool MASK, bool DEFER, int ABL = 0>
 __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
     __shared__ char k_lds[KVB * D * 2];   // [t][d] bf16, swizzled rows
@@ -129,11 +132,16 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
                         vbase + chunk * p.v_sc + tin * p.v_sl + d8 * 8);
                 }
                 *reinterpret_cast<uint4*>(&k_lds[t_local * K_ROW + swz(t_local, d8 * 16)]) = kraw;
-                const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vraw);
+                if (ABL == 1) {  // ablation: linear b128 V write (wrong layout)
+                    *reinterpret_cast<uint4*>(&vt_lds[((t_local * 8 + d8) * 16) %
+                                                      (D * KVB * 2 - 16)]) = vraw;
+                } else {
+                    const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vraw);
 #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int d = d8 * 8 + j;
-                    *reinterpret_cast<uint16_t*>(&vt_lds[d * VT_ROW + swz16(d, t_local * 2)]) = ve[j];
+                    for (int j = 0; j < 8; ++j) {
+                        const int d = d8 * 8 + j;
+                        *reinterpret_cast<uint16_t*>(&vt_lds[d * VT_ROW + swz16(d, t_local * 2)]) = ve[j];
+                    }
                 }
             }
         }
@@ -260,7 +268,15 @@ static void launch_geom(const FlashAttnParams& p, hipStream_t stream) {
         const char* e = std::getenv("DFA_ATTN_DEFER");
         return e == nullptr || e[0] != '0';  // defer-max on by default
     }();
+    static const int abl = [] {
+        const char* e = std::getenv("DFA_ATTN_ABLATE");
+        return e ? atoi(e) : 0;
+    }();
     const bool mask = (p.NC * p.LC) % KVB != 0;
+    if (abl == 1) {  // PMC ablation only — numerics are wrong by design
+        flash_attn_d64_kernel<NW, false, true, 1><<<grid, block, 0, stream>>>(p);
+        return;
+    }
     if (!mask && defer)
         flash_attn_d64_kernel<NW, false, true><<<grid, block, 0, stream>>>(p);
     else if (!mask)
