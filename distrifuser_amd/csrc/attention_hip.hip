@@ -114,7 +114,13 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
     const int n_tiles = (int)((Lkv + KVB - 1) / KVB);
     for (int tile = 0; tile < n_tiles; ++tile) {
         const int64_t t0 = (int64_t)tile * KVB;
-        // ---- cooperative staging: K [t][d], V^T [d][t] ----
+        // ---- cooperative staging ----
+        // K [t][d]: thread (t-row, 16B column) -> b128 LDS write.
+        // V^T [d][t]: lane = d COLUMN, 8 consecutive tokens per pass — the
+        // global reads stay perfectly coalesced (64 lanes x 2B = one 128B row
+        // per token) and the LDS write becomes ONE b128 per lane instead of
+        // 8 scalar ds_write_b16 (the transpose-write conflicts were 15% of
+        // kernel time — see profiles/attention_ladder_r01.md ablation).
         {
             const int tl = tid / 8;                      // covers NW*8 rows/pass
             const int d8 = tid % 8;
@@ -123,27 +129,34 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
             for (int rep = 0; rep < KVB / ROWS_PER_PASS; ++rep) {
                 const int t_local = tl + rep * ROWS_PER_PASS;
                 const int64_t t_glob = t0 + t_local;
-                uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
+                uint4 kraw = {0, 0, 0, 0};
                 if (!MASK || t_glob < Lkv) {
                     const int64_t chunk = t_glob / p.LC;
                     const int64_t tin = t_glob % p.LC;
                     kraw = *reinterpret_cast<const uint4*>(
                         kbase + chunk * p.k_sc + tin * p.k_sl + d8 * 8);
-                    vraw = *reinterpret_cast<const uint4*>(
-                        vbase + chunk * p.v_sc + tin * p.v_sl + d8 * 8);
                 }
                 *reinterpret_cast<uint4*>(&k_lds[t_local * K_ROW + swz(t_local, d8 * 16)]) = kraw;
-                if (ABL == 1) {  // ablation: linear b128 V write (wrong layout)
-                    *reinterpret_cast<uint4*>(&vt_lds[((t_local * 8 + d8) * 16) %
-                                                      (D * KVB * 2 - 16)]) = vraw;
-                } else {
-                    const uint16_t* ve = reinterpret_cast<const uint16_t*>(&vraw);
+            }
+            // V^T: wave w stages token rows [w*8, w*8+8) of each 8*NW-row pass
+            constexpr int VROWS_PER_PASS = NW * 8;
+            const int d = lane;  // this lane's V^T row (d column of V)
 #pragma unroll
-                    for (int j = 0; j < 8; ++j) {
-                        const int d = d8 * 8 + j;
-                        *reinterpret_cast<uint16_t*>(&vt_lds[d * VT_ROW + swz16(d, t_local * 2)]) = ve[j];
-                    }
+            for (int rep = 0; rep < KVB / VROWS_PER_PASS; ++rep) {
+                const int tb_local = rep * VROWS_PER_PASS + wave * 8;
+                const int64_t tb_glob = t0 + tb_local;
+                uint16_t ve[8];
+                // tb_glob is a multiple of 8 and LC % 8 == 0 (host-checked),
+                // so the 8-token window lies in one chunk
+                const int64_t chunk = tb_glob / p.LC;
+                const int64_t tin = tb_glob % p.LC;
+                const uint16_t* vp = vbase + chunk * p.v_sc + tin * p.v_sl + d;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    ve[j] = (!MASK || tb_glob + j < Lkv) ? vp[j * p.v_sl] : (uint16_t)0;
                 }
+                *reinterpret_cast<uint4*>(&vt_lds[d * VT_ROW + swz16(d, tb_local * 2)]) =
+                    *reinterpret_cast<const uint4*>(ve);
             }
         }
         __syncthreads();
@@ -269,15 +282,7 @@ static void launch_geom(const FlashAttnParams& p, hipStream_t stream) {
         const char* e = std::getenv("DFA_ATTN_DEFER");
         return e == nullptr || e[0] != '0';  // defer-max on by default
     }();
-    static const int abl = [] {
-        const char* e = std::getenv("DFA_ATTN_ABLATE");
-        return e ? atoi(e) : 0;
-    }();
     const bool mask = (p.NC * p.LC) % KVB != 0;
-    if (abl == 1) {  // PMC ablation only — numerics are wrong by design
-       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, false, true, 1>), dim3(grid), dim3(block), 0, stream, p);
-        return;
-    }
     if (!mask && defer)
        hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, false, true>), dim3(grid), dim3(block), 0, stream, p);
     else if (!mask)

@@ -148,6 +148,9 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
             LC = t.size(3);
             sc = t.stride(2);
             sl = t.stride(3);
+            // V staging reads 8-token windows assuming they never straddle a
+            // chunk boundary
+            TORCH_CHECK(NC == 1 || LC % 8 == 0, "chunk length must be a multiple of 8");
         }
         TORCH_CHECK(sl % 8 == 0 && sc % 8 == 0 && sh % 8 == 0 && sb % 8 == 0,
                     "KV strides must be 16B-aligned (multiples of 8 elements)");

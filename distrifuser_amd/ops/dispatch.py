@@ -60,6 +60,8 @@ def _flash_ok(q: torch.Tensor, *kv: torch.Tensor) -> bool:
             return False
         if any(s % 8 != 0 for s in t.stride()[:-1]):
             return False
+        if t.dim() == 5 and t.shape[2] > 1 and t.shape[3] % 8 != 0:
+            return False  # chunked KV: 8-token windows must not straddle chunks
     return True
 
 
