@@ -145,6 +145,12 @@ class PatchConv2d(nn.Module):
         k = conv.kernel_size[0]
         s = conv.stride[0]
         pw = conv.padding[1]
+        # stride-2 halo convs need an even local band, otherwise the per-rank
+        # output rows would not tile the global output (choose height divisible
+        # by 8 * 2 * n_device_per_batch for SDXL's two downsamples)
+        assert s == 1 or x.shape[2] % 2 == 0, (
+            f"patch band of {x.shape[2]} rows cannot be downsampled evenly"
+        )
         if k != 3 or conv.padding[0] != 1 or s not in (1, 2):
             parts = [p for p in (top, x, bot) if p is not None]
             pad_top = conv.padding[0] if top is None else 0
