@@ -69,10 +69,18 @@ def _chunked_single_head_attention(
     (4096*230k*4B ~= 3.8 GB) — bounded regardless of resolution.
     """
     scale = q.shape[-1] ** -0.5
+    # GPU: bf16 GEMMs (hipBLASLt) with fp32 softmax — at 230k tokens the
+    # score GEMM is ~5e16 FLOP and fp32 (vector-ALU only on CDNA4, no fp32
+    # MFMA) would take minutes; CPU keeps fp32 for the test oracle.
+    mm_dtype = torch.bfloat16 if q.is_cuda else torch.float32
+    qs = (q.float() * scale).to(mm_dtype)
+    ks = k.to(mm_dtype)
+    vs = v.to(mm_dtype)
     outs = []
     for s in range(0, q.shape[1], chunk):
-        scores = torch.einsum("bqc,bkc->bqk", q[:, s : s + chunk].float() * scale, k.float())
-        outs.append(torch.einsum("bqk,bkc->bqc", scores.softmax(dim=-1), v.float()))
+        scores = torch.einsum("bqc,bkc->bqk", qs[:, s : s + chunk], ks)
+        probs = scores.float().softmax(dim=-1).to(mm_dtype)
+        outs.append(torch.einsum("bqk,bkc->bqc", probs, vs))
     return torch.cat(outs, dim=1).to(q.dtype)
 
 
