@@ -32,6 +32,12 @@ OPEN_CLIP_BIG_G = CLIPTextConfig(
     hidden_size=1280, intermediate_size=5120, num_layers=32, num_heads=20,
     projection_dim=1280, act="gelu",
 )
+# SD2.x text encoder (OpenCLIP ViT-H text tower, truncated to 23 layers as in
+# the stabilityai/stable-diffusion-2-1 checkpoint)
+OPEN_CLIP_VIT_H = CLIPTextConfig(
+    hidden_size=1024, intermediate_size=4096, num_layers=23, num_heads=16,
+    act="gelu",
+)
 TINY_CLIP = CLIPTextConfig(
     vocab_size=1000, hidden_size=16, intermediate_size=32, num_layers=2, num_heads=2,
     projection_dim=16, act="gelu",

@@ -71,6 +71,23 @@ SD15_UNET = UNetConfig(
     sample_size=64,
 )
 
+# stabilityai/stable-diffusion-2-1 unet/config.json shapes
+SD21_UNET = UNetConfig(
+    block_out_channels=(320, 640, 1280, 1280),
+    down_block_types=(
+        "CrossAttnDownBlock2D",
+        "CrossAttnDownBlock2D",
+        "CrossAttnDownBlock2D",
+        "DownBlock2D",
+    ),
+    transformer_layers_per_block=(1, 1, 1, 1),
+    num_attention_heads=(5, 10, 20, 20),  # head_dim 64 per block
+    cross_attention_dim=1024,
+    use_linear_projection=True,
+    addition_embed_type=None,
+    sample_size=96,
+)
+
 # Tiny config for CPU tests (structure-preserving, 64x smaller)
 TINY_UNET = UNetConfig(
     block_out_channels=(32, 64),

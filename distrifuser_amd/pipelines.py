@@ -365,6 +365,12 @@ class DistriSDPipeline(_DistriPipelineBase):
             )
             vae_cfg, clip_cfg = TINY_VAE, TINY_CLIP
             tokenizer = SimpleTokenizer(vocab_size=TINY_CLIP.vocab_size)
+        elif preset in ("sd21", "sd2"):
+            from .models.clip import OPEN_CLIP_VIT_H
+            from .models.unet import SD21_UNET
+
+            unet_cfg, vae_cfg, clip_cfg = SD21_UNET, SD_VAE, OPEN_CLIP_VIT_H
+            tokenizer = SimpleTokenizer()
         else:
             unet_cfg, vae_cfg, clip_cfg = SD15_UNET, SD_VAE, CLIP_VIT_L
             tokenizer = SimpleTokenizer()
