@@ -86,21 +86,30 @@ class DistriUNet(nn.Module):
         return sample[tuple(sl)].contiguous(), dim
 
     def _gather_patches(self, local: torch.Tensor, dim: int, b_full: int) -> torch.Tensor:
-        """All-gather per-rank patches over WORLD and reassemble (2,C,H,W)."""
+        """All-gather per-rank patches over WORLD and reassemble (2,C,H,W).
+
+        The gather rides FLAT per-rank buffers so their shapes are stable
+        across steps even when `split_scheme="alternate"` switches the split
+        axis (shape-changing reallocation would invalidate captured
+        hipGraphs; the reference flattened for the same reason,
+        naive_patch_sdxl.py:147-155)."""
         cfg = self.distri_config
         ws = cfg.world_size
         n = cfg.n_device_per_batch
         local = local.contiguous()
-        if self.buffer_list is None or self.buffer_list[0].shape != local.shape:
-            self.buffer_list = [torch.empty_like(local) for _ in range(ws)]
+        numel = local.numel()
+        if self.buffer_list is None or self.buffer_list[0].numel() != numel:
+            flat = torch.empty(ws, numel, device=local.device, dtype=local.dtype)
+            self.buffer_list = [flat[i] for i in range(ws)]
             self.output_buffer = None
-        dist.all_gather(self.buffer_list, local, async_op=False)
+        dist.all_gather(self.buffer_list, local.reshape(-1), async_op=False)
+        views = [b.view(local.shape) for b in self.buffer_list]
         if cfg.split_batch:
-            b0 = torch.cat(self.buffer_list[:n], dim=dim)
-            b1 = torch.cat(self.buffer_list[n:], dim=dim)
+            b0 = torch.cat(views[:n], dim=dim)
+            b1 = torch.cat(views[n:], dim=dim)
             out = torch.cat([b0, b1], dim=0)
         else:
-            out = torch.cat(self.buffer_list, dim=dim)
+            out = torch.cat(views, dim=dim)
         if self.output_buffer is None or self.output_buffer.shape != out.shape:
             self.output_buffer = torch.empty_like(out)
         self.output_buffer.copy_(out)
