@@ -52,6 +52,8 @@ def get_args():
     p.add_argument("--warmup_times", type=int, default=5)
     p.add_argument("--test_times", type=int, default=20)
     p.add_argument("--ignore_ratio", type=float, default=0.2)
+    p.add_argument("--preset", type=str, default="sdxl", choices=["sdxl", "tiny"],
+                   help="tiny = fast CPU smoke of the CLI")
     return p.parse_args()
 
 
@@ -72,7 +74,7 @@ def main():
     dtype = torch.bfloat16 if torch.cuda.is_available() else torch.float32
     pipe = DistriSDXLPipeline.from_pretrained(
         cfg, torch_dtype=dtype, scheduler=args.scheduler,
-        pretrained_model_name_or_path=args.pretrained,
+        pretrained_model_name_or_path=args.pretrained, preset=args.preset,
     )
 
     def run(output_type):
