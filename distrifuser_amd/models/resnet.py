@@ -13,7 +13,6 @@ import torch.distributed as dist
 import torch.nn.functional as F
 from torch import nn
 
-from ..parallel.state import ParallelState
 from .layers import LayerFactory
 
 

@@ -10,14 +10,12 @@ checkpoints load 1:1 (models/weights.py).
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
-import torch.nn.functional as F
 from torch import nn
 
 from ..parallel.state import ParallelState
-from ..utils.config import DistriConfig
 from .embeddings import TimestepEmbedding, sinusoidal_embedding
 from .layers import LayerFactory
 from .resnet import Downsample2D, ResnetBlock2D, Upsample2D

@@ -20,7 +20,7 @@ from .models.clip import CLIP_VIT_L, OPEN_CLIP_BIG_G, TINY_CLIP, CLIPTextEncoder
 from .models.distri_unet import DistriUNet
 from .models.tokenizer import SimpleTokenizer
 from .models.unet import SD15_UNET, SDXL_UNET, TINY_UNET, UNetConfig
-from .models.vae import SD_VAE, SDXL_VAE, TINY_VAE, VAEDecoder, VAEDecoderConfig
+from .models.vae import SD_VAE, SDXL_VAE, TINY_VAE, VAEDecoder
 from .models import weights as weight_io
 from .schedulers import get_scheduler
 from .utils.comm import PatchParallelismCommManager
