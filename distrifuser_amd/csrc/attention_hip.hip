@@ -161,10 +161,6 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
         }
         __syncthreads();
 
-        // uniform per-tile tail flag: full tiles skip the per-element
-        // mask even in the MASK instantiation (Lkv % KVB != 0 shapes,
-        // e.g. SDXL's 14400-token stage, were paying 3 VALU ops/element)
-        const bool tile_tail = MASK && (t0 + KVB > Lkv);
 #pragma unroll
         for (int st = 0; st < KVB / 32; ++st) {  // 32-token sub-tiles
             // ---- S^T[kv32][q32] = K_sub x Q^T ----
@@ -184,7 +180,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 float v = s[r];
-                if (MASK && tile_tail) {
+                if (MASK) {
                     const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
                     if (t0 + st * 32 + crow >= Lkv) v = -1e30f;
                 }
