@@ -35,10 +35,11 @@ constexpr int QW = 32;     // q rows per wave
 // waves per block is a template parameter: bigger Q blocks amortize the KV
 // stream (each KV tile is re-read Lq/QBLK times), smaller blocks keep small
 // Lq shapes filled. Measured: 4w->8w at L=57.6k was +53% (439->673 TF).
-constexpr int KVB = 128;   // kv tokens per LDS tile
+// KVB (kv tokens per LDS tile) is a template parameter: 128 amortizes the
+// staging barriers; 64 keeps shapes with Lkv % 128 == 64 (SDXL's 14400-token
+// stage) on the mask-free fast path.
 constexpr int D = 64;
 constexpr int K_ROW = D * 2;    // K LDS row bytes   [t][d]
-constexpr int VT_ROW = KVB * 2; // V^T LDS row bytes [d][t]
 
 typedef float float4v_ __attribute__((ext_vector_type(4)));
 typedef float float16v __attribute__((ext_vector_type(16)));
@@ -49,10 +50,11 @@ __device__ __forceinline__ int swz(int row, int byte_off) {
     return byte_off ^ ((row & 7) << 4);
 }
 
-__device__ __forceinline__ int swz16(int row, int byte_off) {
-    // 16-window XOR within a 256 B row (V^T tile): spreads the 32-row
-    // column reads 2-way, which is free on the 32-bank LDS (guide m136)
-    return byte_off ^ ((row & 15) << 4);
+template <int ROW_BYTES>
+__device__ __forceinline__ int vt_swz(int row, int byte_off) {
+    // XOR over all 16 B windows of the V^T row: 16 windows at 256 B rows
+    // (2-way conflicts, free per guide m136), 8 at 128 B rows
+    return byte_off ^ ((row & (ROW_BYTES / 16 - 1)) << 4);
 }
 
 __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
@@ -64,9 +66,10 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
 // ABL: perf-ablation variants (wrong numerics, PMC probing only; selected
 // by DFA_ATTN_ABLATE): 1 = V^T staging as a single linear conflict-free
 // b128 write instead of the 8-lane scalar transpose.
-template <int NW, bool MASK, bool DEFER, int ABL = 0>
+template <int NW, int KVB, bool MASK, bool DEFER>
 __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
+    constexpr int VT_ROW = KVB * 2;  // V^T LDS row bytes [d][t]
     __shared__ char k_lds[KVB * D * 2];   // [t][d] bf16, swizzled rows
     __shared__ char vt_lds[D * KVB * 2];  // [d][t] bf16, swizzled rows
 
@@ -155,7 +158,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
                 for (int j = 0; j < 8; ++j) {
                     ve[j] = (!MASK || tb_glob + j < Lkv) ? vp[j * p.v_sl] : (uint16_t)0;
                 }
-                *reinterpret_cast<uint4*>(&vt_lds[d * VT_ROW + swz16(d, tb_local * 2)]) =
+                *reinterpret_cast<uint4*>(&vt_lds[d * VT_ROW + vt_swz<VT_ROW>(d, tb_local * 2)]) =
                     *reinterpret_cast<const uint4*>(ve);
             }
         }
@@ -246,7 +249,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
                 for (int kt = 0; kt < 2; ++kt) {
                     const int d = dt * 32 + lo;
                     short8 vf = *reinterpret_cast<const short8*>(
-                        &vt_lds[d * VT_ROW + swz16(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
+                        &vt_lds[d * VT_ROW + vt_swz<VT_ROW>(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
                     ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
                 }
             }
@@ -273,8 +276,8 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
 
 #include <cstdlib>
 
-template <int NW>
-static void launch_geom(const FlashAttnParams& p, hipStream_t stream) {
+template <int NW, int KVB, bool MASK>
+static void launch_var(const FlashAttnParams& p, hipStream_t stream) {
     const int qblk = QW * NW;
     dim3 grid((unsigned)((p.Lq + qblk - 1) / qblk), (unsigned)(p.B * p.H));
     dim3 block(NW * WAVE_SIZE);
@@ -282,23 +285,24 @@ static void launch_geom(const FlashAttnParams& p, hipStream_t stream) {
         const char* e = std::getenv("DFA_ATTN_DEFER");
         return e == nullptr || e[0] != '0';  // defer-max on by default
     }();
-    const bool mask = (p.NC * p.LC) % KVB != 0;
-    if (!mask && defer)
-       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, false, true>), dim3(grid), dim3(block), 0, stream, p);
-    else if (!mask)
-       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, false, false>), dim3(grid), dim3(block), 0, stream, p);
-    else if (defer)
-       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, true, true>), dim3(grid), dim3(block), 0, stream, p);
+    if (defer)
+       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, KVB, MASK, true>), dim3(grid), dim3(block), 0, stream, p);
     else
-       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, true, false>), dim3(grid), dim3(block), 0, stream, p);
+       hipLaunchKernelGGL(( flash_attn_d64_kernel<NW, KVB, MASK, false>), dim3(grid), dim3(block), 0, stream, p);
 }
 
 void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
     // 8-wave (256-row) blocks measured best across the SD-family shapes
-    // (16-wave was neutral at L=57.6k: KV re-reads stop being the bound and
-    // the 16-wave barrier domain eats the gain). The geometry template stays
-    // for future tuning.
-    launch_geom<8>(p, stream);
+    // (16-wave was neutral at L=57.6k). KVB: prefer 128-token tiles when the
+    // sequence tiles evenly, else 64 (still mask-free for Lkv % 64 == 0,
+    // e.g. SDXL's 14400-token stage), else the masked variant.
+    const int64_t Lkv = p.NC * p.LC;
+    if (Lkv % 128 == 0)
+        launch_var<8, 128, false>(p, stream);
+    else if (Lkv % 64 == 0)
+        launch_var<8, 64, false>(p, stream);
+    else
+        launch_var<8, 128, true>(p, stream);
 }
 
 // ---- fragment-layout probes (tests/test_ops_gpu.py) ------------------------
