@@ -99,6 +99,21 @@ at::Tensor group_norm_silu(const at::Tensor& x_, int64_t num_groups,
     return y;
 }
 
+at::Tensor gn_merge_stats(const at::Tensor& buffer, int64_t slot_off, int64_t own,
+                          const at::Tensor& fresh_, bool corrected) {
+    // buffer: [n_peers, row_stride] (the flat comm buffer); fresh: [2, N, G,...]
+    TORCH_CHECK(buffer.is_cuda() && buffer.dim() == 2 && buffer.stride(1) == 1);
+    auto fresh = fresh_.contiguous();
+    const int ng = (int)(fresh.numel() / 2);
+    TORCH_CHECK(fresh.scalar_type() == buffer.scalar_type());
+    auto out = at::empty({2, (long)ng}, buffer.options().dtype(at::kFloat));
+    launch_gn_merge_stats(buffer.data_ptr(), buffer.stride(0), slot_off,
+                          (int)buffer.size(0), (int)own, fresh.data_ptr(),
+                          out.data_ptr<float>(), ng, corrected, dtype_of(buffer),
+                          cur_stream());
+    return out;
+}
+
 at::Tensor geglu(const at::Tensor& h_) {
     TORCH_CHECK(h_.is_cuda());
     auto h = h_.contiguous();
@@ -200,6 +215,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("group_norm_apply", &group_norm_apply, "normalize+affine(+SiLU) from given moments");
     m.def("group_norm_silu", &group_norm_silu, "fused GroupNorm(+SiLU)");
     m.def("geglu", &geglu, "a * gelu(gate) over last-dim halves");
+    m.def("gn_merge_stats", &gn_merge_stats,
+          "merge stale peer GN moments with fresh local ones (one launch)");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
     m.def("mfma_probe", &mfma_probe, "dump mfma_f32_16x16x32_bf16 fragment mapping");
     m.def("mfma_probe32", &mfma_probe32, "dump mfma_f32_32x32x16_bf16 fragment mapping");

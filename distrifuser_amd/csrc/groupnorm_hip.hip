@@ -151,7 +151,91 @@ void gn_apply_t(const void* x, void* y, const float* mean, const float* meansq, 
     }
 }
 
+template <typename T, bool CORRECTED>
+__global__ void gn_merge_stats_kernel(T* __restrict__ buffer, int64_t row_stride,
+                                      int64_t slot_off, int n_peers, int own,
+                                      const T* __restrict__ fresh, float* __restrict__ out,
+                                      int ng) {
+    const int g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= ng) return;
+    // moments live as [mean[0..ng), meansq[ng..2ng)] within each peer slot
+    float m_sum = 0.f, q_sum = 0.f;
+    float m_own = 0.f, q_own = 0.f;
+    for (int p = 0; p < n_peers; ++p) {
+        const T* slot = buffer + p * row_stride + slot_off;
+        float m = to_f32(slot[g]);
+        float q = to_f32(slot[ng + g]);
+        if (p == own) {
+            m_own = m;
+            q_own = q;
+            if (!CORRECTED) {  // stale_gn substitutes the fresh slot
+                m = to_f32(fresh[g]);
+                q = to_f32(fresh[ng + g]);
+            }
+        }
+        m_sum += m;
+        q_sum += q;
+    }
+    float mean = m_sum / n_peers;
+    float msq = q_sum / n_peers;
+    const float f_m = to_f32(fresh[g]);
+    const float f_q = to_f32(fresh[ng + g]);
+    if (CORRECTED) {
+        mean += f_m - m_own;
+        msq += f_q - q_own;
+        // negative-variance guard: fall back to the local slice variance
+        // (reference pp/groupnorm.py:60-63)
+        const float var = msq - mean * mean;
+        if (var < 0.f) msq = mean * mean + (f_q - f_m * f_m);
+    }
+    out[g] = mean;
+    out[ng + g] = msq;
+    // stage fresh into our slot for the async all-gather
+    T* own_slot = buffer + own * row_stride + slot_off;
+    own_slot[g] = fresh[g];
+    own_slot[ng + g] = fresh[ng + g];
+}
+
 }  // namespace
+
+void launch_gn_merge_stats(void* buffer, int64_t row_stride, int64_t slot_off, int n_peers,
+                           int own, const void* fresh, float* out, int ng, bool corrected,
+                           int dtype, hipStream_t stream) {
+    const int block = 256;
+    const int grid = (ng + block - 1) / block;
+    switch (dtype) {
+        case DFA_BF16:
+            if (corrected)
+               hipLaunchKernelGGL(( gn_merge_stats_kernel<bf16_t, true>), dim3(grid), dim3(block), 0, stream, 
+                    (bf16_t*)buffer, row_stride, slot_off, n_peers, own, (const bf16_t*)fresh,
+                    out, ng);
+            else
+               hipLaunchKernelGGL(( gn_merge_stats_kernel<bf16_t, false>), dim3(grid), dim3(block), 0, stream, 
+                    (bf16_t*)buffer, row_stride, slot_off, n_peers, own, (const bf16_t*)fresh,
+                    out, ng);
+            break;
+        case DFA_F16:
+            if (corrected)
+               hipLaunchKernelGGL(( gn_merge_stats_kernel<f16_t, true>), dim3(grid), dim3(block), 0, stream, 
+                    (f16_t*)buffer, row_stride, slot_off, n_peers, own, (const f16_t*)fresh,
+                    out, ng);
+            else
+               hipLaunchKernelGGL(( gn_merge_stats_kernel<f16_t, false>), dim3(grid), dim3(block), 0, stream, 
+                    (f16_t*)buffer, row_stride, slot_off, n_peers, own, (const f16_t*)fresh,
+                    out, ng);
+            break;
+        default:
+            if (corrected)
+               hipLaunchKernelGGL(( gn_merge_stats_kernel<float, true>), dim3(grid), dim3(block), 0, stream, 
+                    (float*)buffer, row_stride, slot_off, n_peers, own, (const float*)fresh,
+                    out, ng);
+            else
+               hipLaunchKernelGGL(( gn_merge_stats_kernel<float, false>), dim3(grid), dim3(block), 0, stream, 
+                    (float*)buffer, row_stride, slot_off, n_peers, own, (const float*)fresh,
+                    out, ng);
+            break;
+    }
+}
 
 void launch_gn_stats_partial(const void* x, float* partial, int64_t group_len, int ngt, int dtype,
                              hipStream_t stream) {

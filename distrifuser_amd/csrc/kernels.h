@@ -18,6 +18,19 @@ void launch_gn_apply(const void* x, void* y, const float* mean, const float* mea
                      const void* weight, const void* bias, float eps, int64_t hw,
                      int C, int G, int N, bool silu, int dtype, hipStream_t stream);
 
+// ---- displaced-GN stat merge ------------------------------------------------
+// Merge per-peer stale GroupNorm moments with this rank's fresh ones in ONE
+// launch (replaces a ~10-kernel torch composition per GN layer):
+//   corrected: full = mean(stale) + (fresh - stale_own)   [+ neg-var guard]
+//   stale:     full = mean(stale with own slot := fresh)
+// Also stages `fresh` into the rank's buffer slot (the enqueue needs it).
+// buffer: bf16, n_peers rows of row_stride elements, moment block at
+// buffer[row * row_stride + slot_off .. + 2*ng); fresh: bf16 [2*ng];
+// out: fp32 [2*ng].
+void launch_gn_merge_stats(void* buffer, int64_t row_stride, int64_t slot_off,
+                           int n_peers, int own, const void* fresh, float* out,
+                           int ng, bool corrected, int dtype, hipStream_t stream);
+
 // ---- GEGLU ------------------------------------------------------------------
 // in: [rows, 2*inner]; out: [rows, inner]; out = a * gelu(gate).
 void launch_geglu(const void* in, void* out, int64_t rows, int64_t inner, int dtype,

@@ -23,6 +23,8 @@ Every module shares one ``ParallelState`` (counter / comm manager / config).
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.distributed as dist
 import torch.nn.functional as F
@@ -254,6 +256,22 @@ class PatchGroupNorm(nn.Module):
 
         comm.wait(self._idx)
         fresh = ops.group_norm_stats(x, self.num_groups)  # [2, N, G, 1, 1, 1]
+
+        if not state.in_warmup and x.is_cuda and not os.environ.get("DFA_FORCE_EAGER") == "1":
+            # steady state on GPU: ONE fused kernel merges the stale peer
+            # moments with the fresh local ones (corrected or substitute),
+            # applies the negative-variance guard, and stages fresh into our
+            # buffer slot — replaces a ~10-launch torch composition, which
+            # matters when 8-rank steps are launch-bound
+            moments = ops.hip_ext().gn_merge_stats(
+                comm.buffer, comm.starts[self._idx], cfg.split_idx(), fresh,
+                mode == "corrected_async_gn",
+            )
+            comm.enqueue(self._idx)  # fresh already staged in-slot
+            return ops.group_norm_apply(
+                x, moments[0], moments[1], self.weight, self.bias, self.eps,
+                silu=self.fuse_silu,
+            )
 
         if state.in_warmup:
             dist.all_gather(self._buffer_list, fresh, group=cfg.batch_group)

@@ -194,6 +194,43 @@ def test_group_norm_silu_fused_gpu():
 
 
 @requires_gpu
+def test_gn_merge_stats_gpu():
+    """Fused stale/fresh GN moment merge vs the eager composition
+    (corrected estimator + negative-variance guard + in-slot staging)."""
+    torch.manual_seed(0)
+    dev = _dev()
+    n, N, G = 4, 1, 32
+    ng = N * G
+    row = 2 * ng + 24
+    buffer = torch.rand(n, row, device=dev, dtype=torch.bfloat16) + 0.5
+    slot_off = 8
+    own = 2
+    fresh = (torch.rand(2, N, G, 1, 1, 1, device=dev, dtype=torch.bfloat16) + 0.5)
+    buf0 = buffer.clone()
+
+    for corrected in (True, False):
+        buffer.copy_(buf0)
+        out = hip_ext().gn_merge_stats(buffer, slot_off, own, fresh, corrected).cpu().float()
+        stale = buf0[:, slot_off : slot_off + 2 * ng].view(n, 2, ng).cpu().float()
+        f = fresh.view(2, ng).cpu().float()
+        if corrected:
+            full = stale.mean(0) + (f - stale[own])
+            var = full[1] - full[0] ** 2
+            local_var = f[1] - f[0] ** 2
+            full[1] = torch.where(var < 0, full[0] ** 2 + local_var, full[1])
+        else:
+            sub = stale.clone()
+            sub[own] = f
+            full = sub.mean(0)
+        assert torch.allclose(out, full, atol=2e-2), (
+            f"corrected={corrected} max err {(out - full).abs().max()}"
+        )
+        # fresh staged into own slot
+        staged = buffer[own, slot_off : slot_off + 2 * ng].view(2, ng).cpu()
+        assert torch.equal(staged, fresh.view(2, ng).cpu())
+
+
+@requires_gpu
 def test_geglu_gpu():
     torch.manual_seed(0)
     for shape in [(2, 128, 256), (1, 77, 48), (3, 5, 10)]:
