@@ -2,9 +2,11 @@
 // Flash-attention forward, gfx950, bf16, head_dim 64 (K1/K2/K8 of SURVEY
 // §2.4a): rectangular local-query x global-(stale)-KV attention.
 //
-// v2 structure (cdna_hip_programming.md §B "8-warp 32x32 ladder"):
-// * 4-wave workgroup per (batch, head, 128-row Q tile); each wave owns 32 Q
-//   rows held in registers (4 k-slice A/B fragments).
+// Structure (cdna_hip_programming.md §B "8-warp 32x32 ladder"; evolution and
+// per-step measurements in profiles/attention_ladder_r01.md):
+// * 8-wave workgroup per (batch, head, 256-row Q tile); each wave owns 32 Q
+//   rows held in registers (4 k-slice A/B fragments). Q-block size governs
+//   KV re-read traffic (each KV tile is re-read Lq/QBLK times).
 // * KV streamed in 64-token LDS tiles shared by the waves; K row-major
 //   [64][64], V TRANSPOSED [d][t]; XOR swizzle byte^=((row&7)<<4) breaks the
 //   16-way bank conflict of row-major [.][64] bf16 tiles (guide §6 G4).
