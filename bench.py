@@ -100,6 +100,8 @@ def main() -> None:
         latent_in = torch.cat([latents] * 2)
         latent_in = sched.scale_model_input(latent_in, t)
         noise = pipe.unet(latent_in, t.to(cfg.device) if torch.is_tensor(t) else t, embeds, added)
+        if hasattr(sched, "guided_step"):
+            return sched.guided_step(noise, t, latents, args.guidance_scale)
         nu, nc = noise.chunk(2)
         noise = nu + args.guidance_scale * (nc - nu)
         return sched.step(noise, t, latents)

@@ -120,10 +120,13 @@ class _DistriPipelineBase:
                     prompt_embeds,
                     added_cond_kwargs,
                 )
-                if do_cfg:
-                    n_uncond, n_cond = noise.chunk(2)
-                    noise = n_uncond + guidance_scale * (n_cond - n_uncond)
-                latents = self.scheduler.step(noise, t, latents)
+                if do_cfg and hasattr(self.scheduler, "guided_step"):
+                    latents = self.scheduler.guided_step(noise, t, latents, guidance_scale)
+                else:
+                    if do_cfg:
+                        n_uncond, n_cond = noise.chunk(2)
+                        noise = n_uncond + guidance_scale * (n_cond - n_uncond)
+                    latents = self.scheduler.step(noise, t, latents)
         return latents
 
     def _decode(self, latents: torch.Tensor, output_type: str):

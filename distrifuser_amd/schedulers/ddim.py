@@ -27,3 +27,24 @@ class DDIMScheduler(SchedulerBase):
         direction = (1 - alpha_prev).sqrt() * eps
         prev = alpha_prev.sqrt() * pred_x0 + direction
         return prev.to(sample.dtype)
+
+    def guided_step(self, noise: torch.Tensor, timestep, sample: torch.Tensor,
+                    guidance_scale: float) -> torch.Tensor:
+        # CFG combine + DDIM update; noise is the [uncond; cond] pair. On GPU
+        # this is ONE fused kernel (csrc/scheduler.hip) - the step math runs
+        # outside the captured graphs, so its torch-op launch overhead is
+        # otherwise exposed every denoise step.
+        import os
+
+        t = int(timestep)
+        prev_t = t - self.num_train_timesteps // self.num_inference_steps
+        acp = self.alphas_cumprod
+        alpha_t = float(acp[t])
+        alpha_prev = float(acp[prev_t] if prev_t >= 0 else acp[0])
+        if noise.is_cuda and os.environ.get("DFA_FORCE_EAGER", "0") != "1":
+            from ..ops.dispatch import hip_ext
+
+            return hip_ext().ddim_cfg_step(noise, sample, guidance_scale, alpha_t, alpha_prev)
+        nu, nc = noise.float().chunk(2)
+        eps = nu + guidance_scale * (nc - nu)
+        return self.step(eps, timestep, sample)

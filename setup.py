@@ -22,6 +22,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "bindings.hip"),
         os.path.join(CSRC, "groupnorm.hip"),
         os.path.join(CSRC, "geglu.hip"),
+        os.path.join(CSRC, "scheduler.hip"),
         os.path.join(CSRC, "attention.hip"),
     ],
     extra_compile_args={

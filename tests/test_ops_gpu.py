@@ -241,6 +241,24 @@ def test_geglu_gpu():
 
 
 @requires_gpu
+def test_ddim_cfg_step_gpu():
+    from distrifuser_amd.schedulers import DDIMScheduler
+
+    torch.manual_seed(0)
+    dev = _dev()
+    s = DDIMScheduler()
+    s.set_timesteps(50)
+    noise = torch.randn(2, 4, 64, 64, device=dev, dtype=torch.bfloat16)
+    x = torch.randn(1, 4, 64, 64, device=dev, dtype=torch.bfloat16)
+    t = s.timesteps[7]
+    fused = s.guided_step(noise, t, x, 5.0).float().cpu()
+    nu, nc = noise.float().cpu().chunk(2)
+    eps = nu + 5.0 * (nc - nu)
+    ref = s.step(eps, t, x.float().cpu())
+    assert (fused - ref).abs().max() < 0.03
+
+
+@requires_gpu
 def test_gpu_ops_fail_loudly_without_ext(monkeypatch):
     """On a GPU box, a missing extension must raise, not fall back silently."""
     from distrifuser_amd.ops import dispatch

@@ -77,6 +77,21 @@ def test_euler_scale_model_input():
     assert abs(s.init_noise_sigma - (sigma**2 + 1) ** 0.5) < 1e-6
 
 
+def test_ddim_guided_step_matches_manual():
+    torch.manual_seed(0)
+    s = DDIMScheduler()
+    s.set_timesteps(10)
+    noise = torch.randn(2, 4, 8, 8)
+    x = torch.randn(1, 4, 8, 8)
+    g = 5.0
+    t = s.timesteps[3]
+    fused = s.guided_step(noise, t, x, g)
+    nu, nc = noise.chunk(2)
+    eps = nu + g * (nc - nu)
+    ref = s.step(eps, t, x)
+    assert torch.allclose(fused, ref, atol=1e-6)
+
+
 def test_ddim_monotone_denoise():
     """Variance of the sample should shrink toward the data scale."""
     torch.manual_seed(0)
