@@ -299,7 +299,13 @@ void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
     // sequence tiles evenly, else 64 (still mask-free for Lkv % 64 == 0,
     // e.g. SDXL's 14400-token stage), else the masked variant.
     const int64_t Lkv = p.NC * p.LC;
-    if (Lkv % 128 == 0)
+    static const bool kvb256 = [] {
+        const char* e = std::getenv("DFA_ATTN_KVB256");
+        return e != nullptr && e[0] == '1';  // A/B knob (experimental)
+    }();
+    if (kvb256 && Lkv % 256 == 0 && Lkv >= 4096)
+        launch_var<8, 256, false>(p, stream);
+    else if (Lkv % 128 == 0)
         launch_var<8, 128, false>(p, stream);
     else if (Lkv % 64 == 0)
         launch_var<8, 64, false>(p, stream);
