@@ -36,25 +36,27 @@ constexpr int QW = 32;     // q rows per wave
 // waves per block is a template parameter: bigger Q blocks amortize the KV
 // stream (each KV tile is re-read Lq/QBLK times), smaller blocks keep small
 // Lq shapes filled. Measured: 4w->8w at L=57.6k was +53% (439->673 TF).
-// KVB (kv tokens per LDS tile) is a template parameter: 128 amortizes the
-// staging barriers; 64 keeps shapes with Lkv % 128 == 64 (SDXL's 14400-token
-// stage) on the mask-free fast path.
-constexpr int D = 64;
-constexpr int K_ROW = D * 2;    // K LDS row bytes   [t][d]
+// KVB (kv tokens per LDS tile) and DPAD (head_dim padded to a multiple of
+// 32) are template parameters: KVB=128 amortizes staging barriers, KVB=64
+// keeps Lkv % 128 == 64 shapes (SDXL's 14400-token stage) mask-free; DPAD
+// covers the SD-family head dims (SDXL/SD2: 64; SD1.5: 40->64, 80->96,
+// 160). The real head_dim rides in FlashAttnParams.Dh; padding lanes carry
+// zeros (exact for QK^T scores and O columns < Dh).
+constexpr int pow2ceil(int v) {
+    int p = 1;
+    while (p < v) p <<= 1;
+    return p;
+}
 
 typedef float float4v_ __attribute__((ext_vector_type(4)));
 typedef float float16v __attribute__((ext_vector_type(16)));
 
-__device__ __forceinline__ int swz(int row, int byte_off) {
-    // 8-window XOR within a 128 B row (K tile; the 16 B-aligned b128 floor
-    // for 32-row column reads is 4-way here — see guide §6 G4)
-    return byte_off ^ ((row & 7) << 4);
-}
-
 template <int ROW_BYTES>
 __device__ __forceinline__ int vt_swz(int row, int byte_off) {
-    // XOR over all 16 B windows of the V^T row: 16 windows at 256 B rows
-    // (2-way conflicts, free per guide m136), 8 at 128 B rows
+    // XOR over all 16 B windows of a power-of-two LDS row: spreads same-
+    // column reads across banks (16 windows at 256 B rows is a free 2-way
+    // conflict per guide m136; 8 windows at 128 B rows is the 4-way floor)
+    static_assert((ROW_BYTES & (ROW_BYTES - 1)) == 0);
     return byte_off ^ ((row & (ROW_BYTES / 16 - 1)) << 4);
 }
 
@@ -67,12 +69,15 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
 // ABL: perf-ablation variants (wrong numerics, PMC probing only; selected
 // by DFA_ATTN_ABLATE): 1 = V^T staging as a single linear conflict-free
 // b128 write instead of the 8-lane scalar transpose.
-template <int NW, int KVB, bool MASK, bool DEFER>
-__global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAttnParams p) {
+template <int NW, int DPAD, int KVB, bool MASK, bool DEFER>
+__global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
-    constexpr int VT_ROW = KVB * 2;  // V^T LDS row bytes [d][t]
-    __shared__ char k_lds[KVB * D * 2];   // [t][d] bf16, swizzled rows
-    __shared__ char vt_lds[D * KVB * 2];  // [d][t] bf16, swizzled rows
+    constexpr int VT_ROW = KVB * 2;            // V^T LDS row bytes [d][t]
+    constexpr int K_ROW = pow2ceil(DPAD * 2);  // K LDS row bytes   [t][d]
+    constexpr int KS = DPAD / 16;              // QK^T k-slices
+    constexpr int DT = DPAD / 32;              // PV / O^T d-tiles
+    __shared__ char k_lds[KVB * K_ROW];    // [t][d] bf16, swizzled rows
+    __shared__ char vt_lds[DPAD * VT_ROW];  // [d][t] bf16, swizzled rows
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
@@ -90,15 +95,21 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
     const uint16_t* kbase = p.k + b * p.k_sb + h * p.k_sh;
     const uint16_t* vbase = p.v + b * p.v_sb + h * p.v_sh;
 
-    // ---- Q fragments: qf[ks] = Q[q = q0 + wave*32 + lo][d = ks*16 + hi*8 ..] ----
-    short8 qf[4];
+    // ---- Q fragments: qf[ks] = Q[q = q0 + wave*32 + lo][d = ks*16 + hi*8 ..]
+    // (8-element chunks beyond Dh are zero — Dh % 8 == 0 host-checked) ----
+    short8 qf[KS];
     const int64_t qrow = q0 + wave * QW + lo;
     const bool qvalid = qrow < p.Lq;
     {
         const uint16_t* qp = qbase + (qvalid ? qrow : (p.Lq - 1)) * p.q_sl;
 #pragma unroll
-        for (int ks = 0; ks < 4; ++ks)
-            qf[ks] = *reinterpret_cast<const short8*>(qp + ks * 16 + hi * 8);
+        for (int ks = 0; ks < KS; ++ks) {
+            const int d0 = ks * 16 + hi * 8;
+            if (p.Dh == DPAD || d0 + 8 <= p.Dh)
+                qf[ks] = *reinterpret_cast<const short8*>(qp + d0);
+            else
+                qf[ks] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
     }
 
     // Online softmax bookkeeping (exp2 domain, guide §B):
@@ -111,7 +122,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
     float m_raw = -1e30f;
     float msc = -1e30f;
     float l_run = 0.f;
-    float16v ot[2] = {};  // O^T tiles: [dt] -> rows d = dt*32 + crow(r,hi), col q=lo
+    float16v ot[DT] = {};  // O^T tiles: [dt] -> rows d = dt*32 + crow(r,hi), col q=lo
     const float scale2 = p.scale * 1.44269504088896340736f;
     const float defer_raw = 11.0f / scale2;
 
@@ -126,41 +137,46 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
         // 8 scalar ds_write_b16 (the transpose-write conflicts were 15% of
         // kernel time — see profiles/attention_ladder_r01.md ablation).
         {
-            const int tl = tid / 8;                      // covers NW*8 rows/pass
-            const int d8 = tid % 8;
-            constexpr int ROWS_PER_PASS = NW * WAVE_SIZE / 8;
+            constexpr int KCOLS = DPAD / 8;  // 16 B chunks per K row
+            constexpr int NCHUNK = KVB * KCOLS;
 #pragma unroll
-            for (int rep = 0; rep < KVB / ROWS_PER_PASS; ++rep) {
-                const int t_local = tl + rep * ROWS_PER_PASS;
+            for (int c = tid; c < NCHUNK; c += NW * WAVE_SIZE) {
+                const int t_local = c / KCOLS;
+                const int d8 = c % KCOLS;
                 const int64_t t_glob = t0 + t_local;
                 uint4 kraw = {0, 0, 0, 0};
-                if (!MASK || t_glob < Lkv) {
+                const bool d_ok = (p.Dh == DPAD) || (d8 * 8 + 8 <= p.Dh);
+                if (d_ok && (!MASK || t_glob < Lkv)) {
                     const int64_t chunk = t_glob / p.LC;
                     const int64_t tin = t_glob % p.LC;
                     kraw = *reinterpret_cast<const uint4*>(
                         kbase + chunk * p.k_sc + tin * p.k_sl + d8 * 8);
                 }
-                *reinterpret_cast<uint4*>(&k_lds[t_local * K_ROW + swz(t_local, d8 * 16)]) = kraw;
+                *reinterpret_cast<uint4*>(
+                    &k_lds[t_local * K_ROW + vt_swz<K_ROW>(t_local, d8 * 16)]) = kraw;
             }
             // V^T: wave w stages token rows [w*8, w*8+8) of each 8*NW-row pass
             constexpr int VROWS_PER_PASS = NW * 8;
-            const int d = lane;  // this lane's V^T row (d column of V)
 #pragma unroll
             for (int rep = 0; rep < KVB / VROWS_PER_PASS; ++rep) {
                 const int tb_local = rep * VROWS_PER_PASS + wave * 8;
                 const int64_t tb_glob = t0 + tb_local;
-                uint16_t ve[8];
                 // tb_glob is a multiple of 8 and LC % 8 == 0 (host-checked),
                 // so the 8-token window lies in one chunk
                 const int64_t chunk = tb_glob / p.LC;
                 const int64_t tin = tb_glob % p.LC;
-                const uint16_t* vp = vbase + chunk * p.v_sc + tin * p.v_sl + d;
+                for (int d = lane; d < DPAD; d += WAVE_SIZE) {
+                    uint16_t ve[8];
+                    const uint16_t* vp = vbase + chunk * p.v_sc + tin * p.v_sl + d;
 #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    ve[j] = (!MASK || tb_glob + j < Lkv) ? vp[j * p.v_sl] : (uint16_t)0;
+                    for (int j = 0; j < 8; ++j) {
+                        const bool ok = d < p.Dh && (!MASK || tb_glob + j < Lkv);
+                        ve[j] = ok ? vp[j * p.v_sl] : (uint16_t)0;
+                    }
+                    *reinterpret_cast<uint4*>(
+                        &vt_lds[d * VT_ROW + vt_swz<VT_ROW>(d, tb_local * 2)]) =
+                        *reinterpret_cast<const uint4*>(ve);
                 }
-                *reinterpret_cast<uint4*>(&vt_lds[d * VT_ROW + vt_swz<VT_ROW>(d, tb_local * 2)]) =
-                    *reinterpret_cast<const uint4*>(ve);
             }
         }
         __syncthreads();
@@ -171,10 +187,10 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
             float16v s = {};
             __builtin_amdgcn_s_setprio(1);  // favor the MFMA wave (guide T5)
 #pragma unroll
-            for (int ks = 0; ks < 4; ++ks) {
+            for (int ks = 0; ks < KS; ++ks) {
                 const int t = st * 32 + lo;
                 short8 kfrag = *reinterpret_cast<const short8*>(
-                    &k_lds[t * K_ROW + swz(t, (ks * 16 + hi * 8) * 2)]);
+                    &k_lds[t * K_ROW + vt_swz<K_ROW>(t, (ks * 16 + hi * 8) * 2)]);
                 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
@@ -201,10 +217,9 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
                 msc = msc_new;
                 l_run *= corr;
 #pragma unroll
-                for (int r = 0; r < 16; ++r) {
-                    ot[0][r] *= corr;
-                    ot[1][r] *= corr;
-                }
+                for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+                    for (int r = 0; r < 16; ++r) ot[dt][r] *= corr;
             }
             float tsum = 0.f;
 #pragma unroll
@@ -245,7 +260,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
             // ---- O^T += V^T x P^T ----
             __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-            for (int dt = 0; dt < 2; ++dt) {
+            for (int dt = 0; dt < DT; ++dt) {
 #pragma unroll
                 for (int kt = 0; kt < 2; ++kt) {
                     const int d = dt * 32 + lo;
@@ -259,16 +274,17 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
         __syncthreads();
     }
 
-    // ---- epilogue: O[q][d] = O^T / l ----
+    // ---- epilogue: O[q][d] = O^T / l (only the real head_dim columns) ----
     if (qvalid) {
         const float inv = l_run > 0.f ? 1.f / l_run : 0.f;
-        uint16_t* op = p.o + ((int64_t)b * p.Lq + qrow) * (p.H * D) + (int64_t)h * D;
+        uint16_t* op = p.o + ((int64_t)b * p.Lq + qrow) * (p.H * p.Dh) + (int64_t)h * p.Dh;
 #pragma unroll
-        for (int dt = 0; dt < 2; ++dt)
+        for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-                op[d] = __builtin_bit_cast(uint16_t, __float2bfloat16(ot[dt][r] * inv));
+                if (p.Dh == DPAD || d < p.Dh)
+                    op[d] = __builtin_bit_cast(uint16_t, __float2bfloat16(ot[dt][r] * inv));
             }
     }
 }
@@ -277,39 +293,38 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_d64_kernel(FlashAtt
 
 #include <cstdlib>
 
-template <int NW, int KVB, bool MASK>
+template <int NW, int DPAD, int KVB, bool MASK>
 static void launch_var(const FlashAttnParams& p, hipStream_t stream) {
     const int qblk = QW * NW;
     dim3 grid((unsigned)((p.Lq + qblk - 1) / qblk), (unsigned)(p.B * p.H));
     dim3 block(NW * WAVE_SIZE);
-    static const bool defer = [] {
-        const char* e = std::getenv("DFA_ATTN_DEFER");
-        return e == nullptr || e[0] != '0';  // defer-max on by default
-    }();
-    if (defer)
-        flash_attn_d64_kernel<NW, KVB, MASK, true><<<grid, block, 0, stream>>>(p);
+    flash_attn_kernel<NW, DPAD, KVB, MASK, true><<<grid, block, 0, stream>>>(p);
+}
+
+template <int DPAD>
+static void launch_dpad(const FlashAttnParams& p, hipStream_t stream) {
+    const int64_t Lkv = p.NC * p.LC;
+    if (Lkv % 128 == 0)
+        launch_var<8, DPAD, 128, false>(p, stream);
+    else if (Lkv % 64 == 0)
+        launch_var<8, DPAD, 64, false>(p, stream);
     else
-        flash_attn_d64_kernel<NW, KVB, MASK, false><<<grid, block, 0, stream>>>(p);
+        launch_var<8, DPAD, 128, true>(p, stream);
 }
 
 void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
-    // 8-wave (256-row) blocks measured best across the SD-family shapes
-    // (16-wave was neutral at L=57.6k). KVB: prefer 128-token tiles when the
-    // sequence tiles evenly, else 64 (still mask-free for Lkv % 64 == 0,
-    // e.g. SDXL's 14400-token stage), else the masked variant.
-    const int64_t Lkv = p.NC * p.LC;
-    static const bool kvb256 = [] {
-        const char* e = std::getenv("DFA_ATTN_KVB256");
-        return e != nullptr && e[0] == '1';  // A/B knob (experimental)
-    }();
-    if (kvb256 && Lkv % 256 == 0 && Lkv >= 4096)
-        launch_var<8, 256, false>(p, stream);
-    else if (Lkv % 128 == 0)
-        launch_var<8, 128, false>(p, stream);
-    else if (Lkv % 64 == 0)
-        launch_var<8, 64, false>(p, stream);
+    // 8-wave (256-row) blocks measured best across the SD-family shapes;
+    // KVB prefers 128-token tiles when the sequence tiles evenly, else 64
+    // (still mask-free for Lkv % 64 == 0, e.g. SDXL's 14400-token stage),
+    // else the masked variant. DPAD covers the SD-family head dims.
+    if (p.Dh <= 64)
+        launch_dpad<64>(p, stream);
+    else if (p.Dh <= 96)
+        launch_dpad<96>(p, stream);
+    else if (p.Dh <= 128)
+        launch_dpad<128>(p, stream);
     else
-        launch_var<8, 128, true>(p, stream);
+        launch_dpad<160>(p, stream);
 }
 
 // ---- fragment-layout probes (tests/test_ops_gpu.py) ------------------------

@@ -150,7 +150,9 @@ at::Tensor ddim_cfg_step(const at::Tensor& noise, const at::Tensor& x, double g,
 at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v) {
     // q: [B,H,Lq,64]; k/v: [B,H,Lkv,64] or [B,H,NC,LC,64] (stale-KV chunks)
     TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "flash_attention: bf16 only");
-    TORCH_CHECK(q.dim() == 4 && q.size(-1) == 64, "head_dim must be 64");
+    const int Dh = (int)q.size(-1);
+    TORCH_CHECK(q.dim() == 4 && Dh % 8 == 0 && Dh <= 160,
+                "head_dim must be a multiple of 8 and <= 160");
     TORCH_CHECK(q.stride(-1) == 1 && k.stride(-1) == 1 && v.stride(-1) == 1,
                 "innermost dim must be contiguous");
     const int B = (int)q.size(0), H = (int)q.size(1);
@@ -164,11 +166,13 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
     p.q_sb = q.stride(0);
     p.q_sh = q.stride(1);
     p.q_sl = q.stride(2);
-    p.scale = 0.125f;  // 1/sqrt(64)
+    p.Dh = Dh;
+    p.scale = 1.0f / std::sqrt((float)Dh);
 
     auto set_kv = [&](const at::Tensor& t, const uint16_t*& ptr, int64_t& sb, int64_t& sh,
                       int64_t& sc, int64_t& sl, int64_t& NC, int64_t& LC) {
         TORCH_CHECK(t.scalar_type() == at::kBFloat16);
+        TORCH_CHECK((int)t.size(-1) == Dh, "k/v head_dim mismatch");
         ptr = reinterpret_cast<const uint16_t*>(t.data_ptr());
         sb = t.stride(0);
         sh = t.stride(1);
@@ -196,7 +200,7 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
     TORCH_CHECK(nc2 == p.NC && lc2 == p.LC, "k/v shape mismatch");
     TORCH_CHECK(p.q_sl % 8 == 0 && p.q_sb % 8 == 0 && p.q_sh % 8 == 0);
 
-    auto o = at::empty({(long)B, (long)Lq, (long)H, 64L}, q.options());
+    auto o = at::empty({(long)B, (long)Lq, (long)H, (long)Dh}, q.options());
     p.o = reinterpret_cast<uint16_t*>(o.data_ptr());
     launch_flash_attention_d64(p, cur_stream());
     return o.permute({0, 2, 1, 3});  // logical [B,H,Lq,64]

@@ -41,7 +41,7 @@ void launch_ddim_cfg_step(const void* noise_u, const void* noise_c, const void* 
                           float g, float sqrt_at, float sqrt_1mat, float sqrt_ap,
                           float sqrt_1map, int64_t total, int dtype, hipStream_t stream);
 
-// ---- Flash attention (bf16, head_dim 64) ------------------------------------
+// ---- Flash attention (bf16, SD-family head dims) ----------------------------
 // q: logical [B, H, Lq, 64]; k/v: logical [B, H, NC, LC, 64] (NC stale-KV
 // chunks of LC tokens; NC=1 for plain attention). All strides in ELEMENTS,
 // innermost head_dim contiguous. o: [B, Lq, H, 64] contiguous output.
@@ -51,6 +51,7 @@ struct FlashAttnParams {
     const uint16_t* v;
     uint16_t* o;
     int B, H;
+    int Dh;  // real head_dim (kernel pads to a multiple of 32)
     int64_t Lq, NC, LC;  // Lkv = NC * LC
     int64_t q_sb, q_sh, q_sl;
     int64_t k_sb, k_sh, k_sc, k_sl;

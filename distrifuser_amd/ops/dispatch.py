@@ -52,8 +52,11 @@ def _use_hip(x: torch.Tensor) -> bool:
 # -- public ops --------------------------------------------------------------
 
 
+_FLASH_HEAD_DIMS = (40, 64, 80, 96, 128, 160)  # SD-family head dims
+
+
 def _flash_ok(q: torch.Tensor, *kv: torch.Tensor) -> bool:
-    if q.dtype != torch.bfloat16 or q.shape[-1] != 64:
+    if q.dtype != torch.bfloat16 or q.shape[-1] not in _FLASH_HEAD_DIMS:
         return False
     for t in (q, *kv):
         if t.stride(-1) != 1:

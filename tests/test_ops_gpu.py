@@ -106,6 +106,22 @@ def test_flash_attention_vs_fp32(b, h, lq, lkv):
 
 
 @requires_gpu
+@pytest.mark.parametrize("d", [40, 80, 96, 128, 160])
+@pytest.mark.parametrize("lq,lkv", [(256, 256), (100, 333)])
+def test_flash_attention_head_dims(d, lq, lkv):
+    """SD1.5 head dims (40/80/160) + other padded-DPAD shapes vs fp32."""
+    torch.manual_seed(0)
+    dev = _dev()
+    q = torch.randn(1, 3, lq, d, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(1, 3, lkv, d, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(1, 3, lkv, d, device=dev, dtype=torch.bfloat16)
+    out = ops.flash_attention(q, k, v).float()
+    ref = _attn_ref(q, k, v)
+    err = (out - ref).abs().max().item()
+    assert err < 0.03, f"d={d} lq={lq} lkv={lkv}: max err {err}"
+
+
+@requires_gpu
 def test_flash_attention_chunked_matches_cat():
     """5-D chunked KV (the stale-KV flat-buffer layout) == concatenated KV."""
     torch.manual_seed(0)
