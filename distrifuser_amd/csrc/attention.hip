@@ -73,9 +73,7 @@ template <int NW, int DPAD, int KVB, bool MASK, bool DEFER>
 __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
     constexpr int VT_ROW = KVB * 2;            // V^T LDS row bytes [d][t]
-    // K rows padded to >=256 B where LDS allows: 16 swizzle windows make the
-    // 32-row column b128 reads 2-way instead of the 4-way floor of 128 B rows
-    constexpr int K_ROW = (DPAD * 2 <= 128) ? 256 : pow2ceil(DPAD * 2);
+    constexpr int K_ROW = pow2ceil(DPAD * 2);  // K LDS row bytes   [t][d]
     constexpr int KS = DPAD / 16;              // QK^T k-slices
     constexpr int DT = DPAD / 32;              // PV / O^T d-tiles
     __shared__ char k_lds[KVB * K_ROW];    // [t][d] bf16, swizzled rows
