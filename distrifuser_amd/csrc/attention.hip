@@ -66,9 +66,6 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
     return r;
 }
 
-// ABL: perf-ablation variants (wrong numerics, PMC probing only; selected
-// by DFA_ATTN_ABLATE): 1 = V^T staging as a single linear conflict-free
-// b128 write instead of the 8-lane scalar transpose.
 template <int NW, int DPAD, int KVB, bool MASK, bool DEFER>
 __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
@@ -290,8 +287,6 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
 }
 
 }  // namespace
-
-#include <cstdlib>
 
 template <int NW, int DPAD, int KVB, bool MASK>
 static void launch_var(const FlashAttnParams& p, hipStream_t stream) {
