@@ -50,6 +50,51 @@ def test_sd_tiny_gpu():
 
 
 @requires_gpu
+def test_hip_vs_eager_unet_integration(monkeypatch):
+    """Whole-U-Net output with HIP kernels vs the eager composition
+    (DFA_FORCE_EAGER=1) — catches kernel-integration drift the per-op tests
+    can't see (layout/stride handling through real module wiring)."""
+    import torch
+
+    from distrifuser_amd import DistriConfig
+    from distrifuser_amd.models import DistriUNet
+    from distrifuser_amd.models.unet import UNetConfig
+
+    # d64 heads so the flash kernel path is exercised
+    cfg_model = UNetConfig(
+        block_out_channels=(64, 128),
+        down_block_types=("DownBlock2D", "CrossAttnDownBlock2D"),
+        layers_per_block=1,
+        transformer_layers_per_block=(1, 1),
+        num_attention_heads=(1, 2),
+        cross_attention_dim=64,
+        norm_num_groups=8,
+        use_linear_projection=True,
+        addition_embed_type=None,
+        sample_size=16,
+    )
+    outs = {}
+    for eager in (False, True):
+        if eager:
+            monkeypatch.setenv("DFA_FORCE_EAGER", "1")
+        else:
+            monkeypatch.delenv("DFA_FORCE_EAGER", raising=False)
+        cfg = DistriConfig(height=128, width=128, do_classifier_free_guidance=False,
+                           use_cuda_graph=False, device="cuda:0")
+        torch.manual_seed(0)
+        unet = DistriUNet(cfg_model, cfg).to(device="cuda:0", dtype=torch.bfloat16).eval()
+        x = torch.randn(1, 4, 16, 16, device="cuda:0", dtype=torch.bfloat16,
+                        generator=torch.Generator("cuda:0").manual_seed(1))
+        ehs = torch.randn(1, 7, 64, device="cuda:0", dtype=torch.bfloat16,
+                          generator=torch.Generator("cuda:0").manual_seed(2))
+        with torch.no_grad():
+            unet.set_counter(0)
+            outs[eager] = unet(x, 500.0, ehs, None).float()
+    err = (outs[True] - outs[False]).abs().max().item()
+    assert err < 0.1, f"HIP vs eager U-Net drift: {err}"
+
+
+@requires_gpu
 def test_sdxl_real_unet_one_step():
     """Full-size SDXL U-Net, one denoise step at 1024^2 on cuda:0 (bf16)."""
     from distrifuser_amd import DistriConfig
