@@ -91,6 +91,39 @@ def test_sdxl_pipeline_parallel_matches_single_ws2():
         )
 
 
+def _psnr(a: torch.Tensor, b: torch.Tensor) -> float:
+    mse = float(((a - b) ** 2).mean())
+    if mse == 0:
+        return float("inf")
+    rng = float(b.max() - b.min())
+    import math
+
+    return 10 * math.log10(rng * rng / mse)
+
+
+def test_mode_ladder_quality_vs_full_sync_ws2():
+    """Quantified staleness-mode ladder (the reference's PSNR workflow,
+    README.md:121-144, as an automated assertion): each displaced mode's
+    latents stay within a PSNR floor of full_sync on a real multi-step
+    denoise, and no_sync (zero comm) is the quality baseline."""
+    torch.manual_seed(0)
+    cfg = DistriConfig(height=128, width=128, use_cuda_graph=False, device="cpu", warmup_steps=2)
+    torch.manual_seed(0)
+    pipe = DistriSDXLPipeline.from_pretrained(cfg, preset="tiny", torch_dtype=torch.float32)
+    g = torch.Generator().manual_seed(3)
+    ref = pipe("a scenic mountain", num_inference_steps=6, output_type="latent", generator=g)
+
+    psnrs = {}
+    for mode in ("corrected_async_gn", "stale_gn", "no_sync"):
+        out = run_distributed(2, _pipeline_worker, (mode,))
+        psnrs[mode] = _psnr(out[0], ref)
+    # displaced modes must track the serial output closely at this scale
+    assert psnrs["corrected_async_gn"] > 25, psnrs
+    assert psnrs["stale_gn"] > 25, psnrs
+    # no cross-patch comm at all is allowed to drift further but stays sane
+    assert psnrs["no_sync"] > 10, psnrs
+
+
 def test_sdxl_pipeline_parallel_patch_ws4():
     """ws=4: CFG split x 2 displaced patches; ranks agree, full_sync == 1-proc."""
     torch.manual_seed(0)
