@@ -127,23 +127,19 @@ at::Tensor geglu(const at::Tensor& h_) {
     return out;
 }
 
-at::Tensor ddim_cfg_step(const at::Tensor& noise, const at::Tensor& x, double g,
-                         double alpha_t, double alpha_prev) {
-    // noise: [2, C, H, W] = [uncond; cond]; x: [1, C, H, W]
+at::Tensor cfg_affine_step(const at::Tensor& noise, const at::Tensor& x, double g, double ca,
+                           double cb) {
+    // noise: [2, C, H, W] = [uncond; cond]; x: [1, C, H, W]; out = ca*x + cb*eps
     TORCH_CHECK(noise.is_cuda() && noise.size(0) == 2);
     auto n = noise.contiguous();
     auto xc = x.contiguous();
     TORCH_CHECK(xc.numel() * 2 == n.numel());
     auto out = at::empty_like(xc);
     const int64_t total = xc.numel();
-    const float sat = std::sqrt((float)alpha_t);
-    const float s1mat = std::sqrt(1.f - (float)alpha_t);
-    const float sap = std::sqrt((float)alpha_prev);
-    const float s1map = std::sqrt(1.f - (float)alpha_prev);
-    launch_ddim_cfg_step(n.data_ptr(),
-                         reinterpret_cast<const char*>(n.data_ptr()) + total * n.element_size(),
-                         xc.data_ptr(), out.data_ptr(), (float)g, sat, s1mat, sap, s1map, total,
-                         dtype_of(xc), cur_stream());
+    launch_cfg_affine_step(
+        n.data_ptr(),
+        reinterpret_cast<const char*>(n.data_ptr()) + total * n.element_size(), xc.data_ptr(),
+        out.data_ptr(), (float)g, (float)ca, (float)cb, total, dtype_of(xc), cur_stream());
     return out;
 }
 
@@ -241,7 +237,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("geglu", &geglu, "a * gelu(gate) over last-dim halves");
     m.def("gn_merge_stats", &gn_merge_stats,
           "merge stale peer GN moments with fresh local ones (one launch)");
-    m.def("ddim_cfg_step", &ddim_cfg_step, "fused CFG combine + DDIM (eta=0) update");
+    m.def("cfg_affine_step", &cfg_affine_step, "fused CFG combine + affine scheduler update");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
     m.def("mfma_probe", &mfma_probe, "dump mfma_f32_16x16x32_bf16 fragment mapping");
     m.def("mfma_probe32", &mfma_probe32, "dump mfma_f32_32x32x16_bf16 fragment mapping");

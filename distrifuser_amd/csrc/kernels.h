@@ -36,10 +36,10 @@ void launch_gn_merge_stats(void* buffer, int64_t row_stride, int64_t slot_off,
 void launch_geglu(const void* in, void* out, int64_t rows, int64_t inner, int dtype,
                   hipStream_t stream);
 
-// ---- fused CFG + DDIM step --------------------------------------------------
-void launch_ddim_cfg_step(const void* noise_u, const void* noise_c, const void* x, void* out,
-                          float g, float sqrt_at, float sqrt_1mat, float sqrt_ap,
-                          float sqrt_1map, int64_t total, int dtype, hipStream_t stream);
+// ---- fused CFG + scheduler step (affine: out = ca*x + cb*eps) ---------------
+void launch_cfg_affine_step(const void* noise_u, const void* noise_c, const void* x, void* out,
+                            float g, float ca, float cb, int64_t total, int dtype,
+                            hipStream_t stream);
 
 // ---- Flash attention (bf16, SD-family head dims) ----------------------------
 // q: logical [B, H, Lq, 64]; k/v: logical [B, H, NC, LC, 64] (NC stale-KV

@@ -44,7 +44,11 @@ class DDIMScheduler(SchedulerBase):
         if noise.is_cuda and os.environ.get("DFA_FORCE_EAGER", "0") != "1":
             from ..ops.dispatch import hip_ext
 
-            return hip_ext().ddim_cfg_step(noise, sample, guidance_scale, alpha_t, alpha_prev)
+            # x' = sqrt(a_p) * (x - sqrt(1-a_t) eps)/sqrt(a_t) + sqrt(1-a_p) eps
+            #    = ca * x + cb * eps
+            ca = (alpha_prev / alpha_t) ** 0.5
+            cb = (1 - alpha_prev) ** 0.5 - ca * (1 - alpha_t) ** 0.5
+            return hip_ext().cfg_affine_step(noise, sample, guidance_scale, ca, cb)
         nu, nc = noise.float().chunk(2)
         eps = nu + guidance_scale * (nc - nu)
         return self.step(eps, timestep, sample)

@@ -40,3 +40,19 @@ class EulerDiscreteScheduler(SchedulerBase):
         prev = x + eps * (sigma_next - sigma)
         self._step_index += 1
         return prev.to(sample.dtype)
+
+    def guided_step(self, noise: torch.Tensor, timestep, sample: torch.Tensor,
+                    guidance_scale: float) -> torch.Tensor:
+        # CFG combine + Euler update in ONE fused kernel on GPU (the update is
+        # affine in (x, eps): ca = 1, cb = sigma_next - sigma)
+        import os
+
+        if noise.is_cuda and os.environ.get("DFA_FORCE_EAGER", "0") != "1":
+            from ..ops.dispatch import hip_ext
+
+            cb = float(self.sigmas[self._step_index + 1] - self.sigmas[self._step_index])
+            self._step_index += 1
+            return hip_ext().cfg_affine_step(noise, sample, guidance_scale, 1.0, cb)
+        nu, nc = noise.float().chunk(2)
+        eps = nu + guidance_scale * (nc - nu)
+        return self.step(eps, timestep, sample)

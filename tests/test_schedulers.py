@@ -92,6 +92,22 @@ def test_ddim_guided_step_matches_manual():
     assert torch.allclose(fused, ref, atol=1e-6)
 
 
+def test_euler_guided_step_matches_manual():
+    torch.manual_seed(0)
+    s = EulerDiscreteScheduler()
+    s.set_timesteps(10)
+    noise = torch.randn(2, 4, 8, 8)
+    x = torch.randn(1, 4, 8, 8)
+    t = s.timesteps[0]
+    fused = s.guided_step(noise, t, x, 7.5)
+    s2 = EulerDiscreteScheduler()
+    s2.set_timesteps(10)
+    nu, nc = noise.chunk(2)
+    eps = nu + 7.5 * (nc - nu)
+    ref = s2.step(eps, t, x)
+    assert torch.allclose(fused, ref, atol=1e-6)
+
+
 def test_ddim_monotone_denoise():
     """Variance of the sample should shrink toward the data scale."""
     torch.manual_seed(0)
