@@ -162,6 +162,12 @@ def main() -> None:
                 "cuda_graph": cfg.use_cuda_graph,
             },
         }
+        if pipe.comm_manager is not None and pipe.comm_manager.stats["gathers"]:
+            s = pipe.comm_manager.stats
+            result["config"]["comm"] = {
+                "async_gathers": s["gathers"],
+                "bytes_per_rank": s["bytes"],
+            }
         print(json.dumps(result))
 
 
