@@ -4,17 +4,24 @@ has no untested plumbing."""
 
 import json
 import os
+import socket
 import subprocess
 import sys
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _run_bench(nproc: int):
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", "29617",
+        "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
         os.path.join(REPO, "bench.py"),
         "--gpus", str(nproc), "--steps", "2", "--warmup", "1",
         "--height", "128", "--width", "128", "--preset", "tiny",
