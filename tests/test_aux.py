@@ -1,0 +1,82 @@
+"""Aux-subsystem coverage: race-debug sync mode, tracing, tokenizer,
+dispatch fallbacks."""
+
+import os
+
+import torch
+
+from distrifuser_amd import DistriConfig, PatchParallelismCommManager
+from distrifuser_amd.models.tokenizer import SimpleTokenizer
+from distrifuser_amd.utils.tracing import StepTimer, trace_range
+
+from conftest import run_distributed
+
+
+def _debug_sync_worker(rank, world_size):
+    import importlib
+
+    import distrifuser_amd.utils.comm as comm_mod
+
+    os.environ["DFA_DEBUG_SYNC"] = "1"
+    importlib.reload(comm_mod)
+    try:
+        cfg = DistriConfig(do_classifier_free_guidance=False, device="cpu")
+        comm = comm_mod.PatchParallelismCommManager(cfg)
+        i0 = comm.register_tensor((4,), torch.float32)
+        comm.create_buffer()
+        comm.enqueue(i0, torch.full((4,), float(rank)))
+        comm.communicate()
+        # in debug-sync mode the gather completed eagerly: no handles pending
+        assert comm.handles[i0] is None
+        bl = comm.get_buffer_list(i0)
+        for p in range(world_size):
+            assert bl[p][0].item() == float(p)
+        return True
+    finally:
+        os.environ.pop("DFA_DEBUG_SYNC", None)
+        importlib.reload(comm_mod)
+
+
+def test_debug_sync_mode_ws2():
+    out = run_distributed(2, _debug_sync_worker)
+    assert all(out.values())
+
+
+def test_step_timer():
+    t = StepTimer(enabled=True, sync=False)
+    for _ in range(3):
+        t.start()
+        sum(range(1000))
+        t.stop()
+    s = t.summary()
+    assert s["n"] == 3 and s["mean_ms"] >= 0
+
+
+def test_trace_range_noop_on_cpu():
+    with trace_range("x"):
+        pass  # must not raise without a GPU
+
+
+def test_tokenizer_shape_and_determinism():
+    tok = SimpleTokenizer()
+    ids = tok(["a cat", "a cat", "a dog"])
+    assert ids.shape == (3, 77)
+    assert torch.equal(ids[0], ids[1])
+    assert not torch.equal(ids[0], ids[2])
+    assert ids[0, 0].item() == tok.bos_token_id
+    # eos is the max id so the pooled-EOT argmax finds it
+    assert ids[0].max().item() == tok.eos_token_id
+
+
+def test_flash_dispatch_falls_back_on_unsupported_dims():
+    """CPU path + unsupported head dim must ride the eager reference."""
+    from distrifuser_amd import ops
+
+    q = torch.randn(1, 2, 8, 32)  # d=32 not in the kernel's dim set
+    k = torch.randn(1, 2, 8, 32)
+    v = torch.randn(1, 2, 8, 32)
+    out = ops.flash_attention(q, k, v)
+    import torch.nn.functional as F
+
+    ref = F.scaled_dot_product_attention(q, k, v)
+    assert torch.allclose(out, ref, atol=1e-6)
