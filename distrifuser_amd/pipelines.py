@@ -14,17 +14,49 @@ diffusers-layout directory with safetensors files is loaded when given.
 
 from __future__ import annotations
 
+import json
+import os
+
 import torch
 
 from .models.clip import CLIP_VIT_L, OPEN_CLIP_BIG_G, TINY_CLIP, CLIPTextEncoder
 from .models.distri_unet import DistriUNet
-from .models.tokenizer import SimpleTokenizer
+from .models.tokenizer import CLIPBPETokenizer, SimpleTokenizer, load_clip_tokenizer
 from .models.unet import SD15_UNET, SDXL_UNET, TINY_UNET, UNetConfig
 from .models.vae import SD_VAE, SDXL_VAE, TINY_VAE, VAEDecoder
 from .models import weights as weight_io
 from .schedulers import get_scheduler
 from .utils.comm import PatchParallelismCommManager
 from .utils.config import DistriConfig
+
+
+def _resolve_tokenizer(pretrained, vocab_size: int, subfolder: str = "tokenizer",
+                       pad_with_zero: bool = False, allow_simple: bool = False):
+    """Real-checkpoint tokenizer policy: a checkpoint directory
+    must ship CLIP BPE files (``<subfolder>/vocab.json`` + ``merges.txt``) —
+    hash-token ids against real weights silently break text conditioning.
+    Our own ``save_pretrained`` leaves a SimpleTokenizer marker so random-init
+    round-trips keep working; anything else must opt in explicitly."""
+    if pretrained is None:
+        return SimpleTokenizer(vocab_size=vocab_size)
+    tok = load_clip_tokenizer(pretrained, subfolder, pad_with_zero=pad_with_zero)
+    if tok is not None:
+        return tok
+    marker = os.path.join(pretrained, subfolder, "tokenizer_config.json")
+    marker_simple = False
+    if os.path.isfile(marker):
+        try:
+            with open(marker, encoding="utf-8") as f:
+                marker_simple = json.load(f).get("tokenizer_class") == "SimpleTokenizer"
+        except (OSError, ValueError):
+            pass
+    if marker_simple or allow_simple:
+        return SimpleTokenizer(vocab_size=vocab_size)
+    raise FileNotFoundError(
+        f"checkpoint {pretrained!r} has no {subfolder}/vocab.json+merges.txt: pairing real "
+        "weights with the hash SimpleTokenizer would feed garbage token ids to the text "
+        "encoders (conditioning silently broken). Ship the CLIP BPE files, or pass "
+        "allow_simple_tokenizer=True for random-init/debug checkpoints.")
 
 
 def _maybe_tqdm(iterable, enabled: bool):
@@ -80,7 +112,24 @@ class _DistriPipelineBase:
         for name, model in components.items():
             os.makedirs(os.path.join(out_dir, name), exist_ok=True)
             sd = {k: v.contiguous().cpu() for k, v in export_diffusers_state_dict(model).items()}
-            save_file(sd, os.path.join(out_dir, name, "model.safetensors"))
+            # diffusers repo layout: unet/vae use diffusion_pytorch_model.*,
+            # text encoders use model.* (ADVICE r1 — keeps the export loadable
+            # by upstream diffusers, not just our own from_pretrained).
+            fname = ("diffusion_pytorch_model.safetensors"
+                     if name in ("unet", "vae") else "model.safetensors")
+            save_file(sd, os.path.join(out_dir, name, fname))
+        tokenizers = {"tokenizer": self.tokenizer}
+        if getattr(self, "tokenizer_2", None) is not None:
+            tokenizers["tokenizer_2"] = self.tokenizer_2
+        for sub, tok in tokenizers.items():
+            tdir = os.path.join(out_dir, sub)
+            os.makedirs(tdir, exist_ok=True)
+            with open(os.path.join(tdir, "tokenizer_config.json"), "w", encoding="utf-8") as f:
+                json.dump({"tokenizer_class": type(tok).__name__}, f)
+            if isinstance(tok, CLIPBPETokenizer):
+                import shutil
+                shutil.copy(tok.vocab_path, os.path.join(tdir, "vocab.json"))
+                shutil.copy(tok.merges_path, os.path.join(tdir, "merges.txt"))
 
     @property
     def device(self):
@@ -228,9 +277,11 @@ class _DistriPipelineBase:
 class DistriSDXLPipeline(_DistriPipelineBase):
     is_sdxl = True
 
-    def __init__(self, distri_config, unet, vae, text_encoder, text_encoder_2, scheduler, tokenizer):
+    def __init__(self, distri_config, unet, vae, text_encoder, text_encoder_2, scheduler,
+                 tokenizer, tokenizer_2=None):
         self.text_encoder = text_encoder
         self.text_encoder_2 = text_encoder_2
+        self.tokenizer_2 = tokenizer_2
         super().__init__(distri_config, unet, vae, scheduler, tokenizer)
 
     @staticmethod
@@ -239,16 +290,21 @@ class DistriSDXLPipeline(_DistriPipelineBase):
         torch_dtype = kwargs.pop("torch_dtype", torch.bfloat16)
         scheduler = kwargs.pop("scheduler", "ddim")
         preset = kwargs.pop("preset", "sdxl")
+        allow_simple = kwargs.pop("allow_simple_tokenizer", False)
         device = distri_config.device
 
         if preset == "tiny":
             unet_cfg, vae_cfg = TINY_UNET, TINY_VAE
             clip1_cfg, clip2_cfg = TINY_CLIP, TINY_CLIP
-            tokenizer = SimpleTokenizer(vocab_size=TINY_CLIP.vocab_size)
         else:
             unet_cfg, vae_cfg = SDXL_UNET, SDXL_VAE
             clip1_cfg, clip2_cfg = CLIP_VIT_L, OPEN_CLIP_BIG_G
-            tokenizer = SimpleTokenizer()
+        tokenizer = _resolve_tokenizer(pretrained, clip1_cfg.vocab_size,
+                                       allow_simple=allow_simple)
+        # SDXL's second tokenizer (open-CLIP) pads with 0 ("!"), not eos.
+        tokenizer_2 = _resolve_tokenizer(pretrained, clip2_cfg.vocab_size,
+                                         subfolder="tokenizer_2", pad_with_zero=True,
+                                         allow_simple=allow_simple)
         unet_cfg = kwargs.pop("unet_config", unet_cfg)
 
         unet = DistriUNet(unet_cfg, distri_config)
@@ -268,23 +324,27 @@ class DistriSDXLPipeline(_DistriPipelineBase):
         te1 = te1.to(device=device, dtype=torch_dtype).eval()
         te2 = te2.to(device=device, dtype=torch_dtype).eval()
         return DistriSDXLPipeline(distri_config, unet, vae, te1, te2,
-                                  get_scheduler(scheduler), tokenizer)
+                                  get_scheduler(scheduler), tokenizer,
+                                  tokenizer_2=tokenizer_2)
 
     @torch.no_grad()
     def encode_prompt(self, prompt, negative_prompt=None, do_classifier_free_guidance=True):
         """Returns (prompt_embeds [B,77,2048], pooled [B,1280]) per branch,
         concatenated [uncond; cond] when CFG is on."""
         device = self.distri_config.device
+        tok2 = self.tokenizer_2 if self.tokenizer_2 is not None else self.tokenizer
         ids1 = self.tokenizer(prompt, device=device)
+        ids2 = tok2(prompt, device=device)
         emb1, _ = self.text_encoder(ids1, hidden_state_index=-2)
-        emb2, pooled = self.text_encoder_2(ids1, hidden_state_index=-2)
+        emb2, pooled = self.text_encoder_2(ids2, hidden_state_index=-2)
         embeds = torch.cat([emb1, emb2], dim=-1)
         if not do_classifier_free_guidance:
             return embeds, pooled
         neg = negative_prompt if negative_prompt is not None else ""
         nids = self.tokenizer(neg, device=device)
+        nids2 = tok2(neg, device=device)
         nemb1, _ = self.text_encoder(nids, hidden_state_index=-2)
-        nemb2, npooled = self.text_encoder_2(nids, hidden_state_index=-2)
+        nemb2, npooled = self.text_encoder_2(nids2, hidden_state_index=-2)
         nembeds = torch.cat([nemb1, nemb2], dim=-1)
         return torch.cat([nembeds, embeds]), torch.cat([npooled, pooled])
 
@@ -351,6 +411,7 @@ class DistriSDPipeline(_DistriPipelineBase):
         torch_dtype = kwargs.pop("torch_dtype", torch.bfloat16)
         scheduler = kwargs.pop("scheduler", "ddim")
         preset = kwargs.pop("preset", "sd15")
+        allow_simple = kwargs.pop("allow_simple_tokenizer", False)
         device = distri_config.device
 
         if preset == "tiny":
@@ -367,16 +428,15 @@ class DistriSDPipeline(_DistriPipelineBase):
                 sample_size=8,
             )
             vae_cfg, clip_cfg = TINY_VAE, TINY_CLIP
-            tokenizer = SimpleTokenizer(vocab_size=TINY_CLIP.vocab_size)
         elif preset in ("sd21", "sd2"):
             from .models.clip import OPEN_CLIP_VIT_H
             from .models.unet import SD21_UNET
 
             unet_cfg, vae_cfg, clip_cfg = SD21_UNET, SD_VAE, OPEN_CLIP_VIT_H
-            tokenizer = SimpleTokenizer()
         else:
             unet_cfg, vae_cfg, clip_cfg = SD15_UNET, SD_VAE, CLIP_VIT_L
-            tokenizer = SimpleTokenizer()
+        tokenizer = _resolve_tokenizer(pretrained, clip_cfg.vocab_size,
+                                       allow_simple=allow_simple)
         unet_cfg = kwargs.pop("unet_config", unet_cfg)
 
         unet = DistriUNet(unet_cfg, distri_config)

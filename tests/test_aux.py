@@ -80,3 +80,50 @@ def test_flash_dispatch_falls_back_on_unsupported_dims():
 
     ref = F.scaled_dot_product_attention(q, k, v)
     assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_clip_bpe_tokenizer(tmp_path):
+    """CLIPBPETokenizer applies merges, wraps with bos/eos, pads per variant."""
+    import json
+
+    from distrifuser_amd.models.tokenizer import CLIPBPETokenizer
+
+    # Tiny vocab: single chars + the merged "lo" pair + "w</w>" endings.
+    vocab = {"<|startoftext|>": 9, "<|endoftext|>": 10,
+             "l": 0, "o": 1, "w</w>": 2, "lo": 3, "lo w</w>": 4, "low</w>": 5,
+             "o</w>": 6, "l</w>": 7, "lo</w>": 8}
+    merges = ["#version: 0.2", "l o", "lo w</w>"]
+    (tmp_path / "vocab.json").write_text(json.dumps(vocab))
+    (tmp_path / "merges.txt").write_text("\n".join(merges))
+    tok = CLIPBPETokenizer(str(tmp_path / "vocab.json"), str(tmp_path / "merges.txt"))
+    ids = tok("low")
+    assert ids.shape == (1, 77)
+    row = ids[0].tolist()
+    # bos, merges l+o -> "lo", lo+"w</w>" -> "low</w>" (id 5), eos, eos padding
+    assert row[0] == 9 and row[1] == 5 and row[2] == 10
+    assert row[3] == 10  # eos-padding (CLIP-L convention)
+    # pooled-EOT argmax must land on the REAL eos (first max occurrence)
+    assert ids[0].argmax().item() == 2
+
+    tok0 = CLIPBPETokenizer(str(tmp_path / "vocab.json"), str(tmp_path / "merges.txt"),
+                            pad_with_zero=True)
+    assert tok0("low")[0, 3].item() == 0  # open-CLIP tokenizer_2 pads with 0
+
+    # "lo" alone ends the word: chars l,o -> "l","o</w>"; merge "l o" applies
+    # only to non-terminal o, so stays ["l", "o</w>"]
+    row2 = tok("lo")[0].tolist()
+    assert row2[:4] == [9, 0, 6, 10]
+
+
+def test_pretrained_requires_real_tokenizer(tmp_path):
+    """from_pretrained(real checkpoint dir without BPE files) raises clearly."""
+    import pytest as _pytest
+
+    from distrifuser_amd import DistriSDXLPipeline
+    from distrifuser_amd.utils.config import DistriConfig
+
+    (tmp_path / "unet").mkdir()
+    cfg = DistriConfig(height=64, width=64, use_cuda_graph=False, device="cpu")
+    with _pytest.raises(FileNotFoundError, match="vocab.json"):
+        DistriSDXLPipeline.from_pretrained(
+            cfg, preset="tiny", pretrained_model_name_or_path=str(tmp_path))

@@ -62,7 +62,8 @@ def test_pipeline_save_and_reload(tmp_path):
     torch.manual_seed(0)
     pipe = DistriSDXLPipeline.from_pretrained(cfg, preset="tiny", torch_dtype=torch.float32)
     pipe.save_pretrained(str(tmp_path))
-    assert (tmp_path / "unet" / "model.safetensors").exists()
+    assert (tmp_path / "unet" / "diffusion_pytorch_model.safetensors").exists()
+    assert (tmp_path / "tokenizer" / "tokenizer_config.json").exists()
 
     torch.manual_seed(123)  # different init; weights must come from disk
     pipe2 = DistriSDXLPipeline.from_pretrained(
