@@ -202,6 +202,13 @@ class _DistriPipelineBase:
         cfg = self.distri_config
         unet = self.unet
 
+        if cfg.device.type == "cuda":
+            # MIOpen find mode for whichever convs still ride torch ops: the
+            # immediate-mode default picks im2col paths at large resolutions
+            # (profiles/rocprof_3840_r01.md). Applies to every entry point,
+            # not just bench.py.
+            torch.backends.cudnn.benchmark = True
+
         needs_comm = cfg.parallelism == "patch" and cfg.n_device_per_batch > 1
         wants_graphs = cfg.use_cuda_graph and cfg.device.type == "cuda"
         if not needs_comm and not wants_graphs:
