@@ -202,6 +202,61 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
     return o.permute({0, 2, 1, 3});  // logical [B,H,Lq,64]
 }
 
+at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
+                   const c10::optional<at::Tensor>& bias, int64_t cout, int64_t stride,
+                   const c10::optional<at::Tensor>& top, const c10::optional<at::Tensor>& bot) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kBFloat16,
+                "x must be CUDA bf16 [B,Cin,H,W]");
+    TORCH_CHECK(x.stride(3) == 1 && x.stride(2) == x.size(3), "x rows must be contiguous");
+    TORCH_CHECK(wp.is_cuda() && wp.dim() == 5 && wp.is_contiguous() &&
+                wp.scalar_type() == at::kBFloat16 && wp.size(0) == 9 &&
+                wp.size(3) == 64 && wp.size(4) == 8,
+                "wp must be packed [9][KS][CT][64][8] bf16");
+    TORCH_CHECK(stride == 1 || stride == 2);
+    const int B = (int)x.size(0), Cin = (int)x.size(1), H = (int)x.size(2), W = (int)x.size(3);
+    const int Ho = (H - 1) / (int)stride + 1, Wo = (W - 1) / (int)stride + 1;
+
+    Conv3x3Params p{};
+    p.x = reinterpret_cast<const uint16_t*>(x.data_ptr());
+    p.wp = reinterpret_cast<const uint16_t*>(wp.data_ptr());
+    p.B = B;
+    p.Cin = Cin;
+    p.Cout = (int)cout;
+    p.H = H;
+    p.W = W;
+    p.Ho = Ho;
+    p.Wo = Wo;
+    p.KS = (int)wp.size(1);
+    p.CT = (int)wp.size(2);
+    TORCH_CHECK(p.KS * 16 >= Cin && p.CT * 32 >= cout, "packed weight too small");
+    p.x_sb = x.stride(0);
+    p.x_sc = x.stride(1);
+    at::Tensor bias_c;
+    if (bias.has_value()) {
+        bias_c = bias->to(at::kBFloat16).contiguous();
+        TORCH_CHECK(bias_c.numel() == cout);
+        p.bias = reinterpret_cast<const uint16_t*>(bias_c.data_ptr());
+    }
+    auto set_halo = [&](const c10::optional<at::Tensor>& h, const uint16_t*& ptr, int64_t& sb,
+                        int64_t& sc) {
+        if (!h.has_value()) return;
+        const at::Tensor& t = *h;
+        TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 && t.stride(-1) == 1 &&
+                    t.size(-1) == W && t.size(1) == Cin,
+                    "halo must be bf16 [B,Cin,(1,)W] with contiguous rows");
+        ptr = reinterpret_cast<const uint16_t*>(t.data_ptr());
+        sb = t.stride(0);
+        sc = t.stride(1);
+    };
+    set_halo(top, p.top, p.t_sb, p.t_sc);
+    set_halo(bot, p.bot, p.b_sb, p.b_sc);
+
+    auto o = at::empty({(long)B, (long)cout, (long)Ho, (long)Wo}, x.options());
+    p.o = reinterpret_cast<uint16_t*>(o.data_ptr());
+    launch_conv3x3(p, (int)stride, cur_stream());
+    return o;
+}
+
 std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b);
 
 }  // namespace
@@ -239,6 +294,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "merge stale peer GN moments with fresh local ones (one launch)");
     m.def("cfg_affine_step", &cfg_affine_step, "fused CFG combine + affine scheduler update");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
+    m.def("conv3x3", &conv3x3,
+          "bf16 implicit-GEMM 3x3 conv, stride 1/2, in-place halo rows");
     m.def("mfma_probe", &mfma_probe, "dump mfma_f32_16x16x32_bf16 fragment mapping");
     m.def("mfma_probe32", &mfma_probe32, "dump mfma_f32_32x32x16_bf16 fragment mapping");
 }

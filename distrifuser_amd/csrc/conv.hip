@@ -1,0 +1,197 @@
+// Implicit-GEMM 3x3 convolution, gfx950, bf16, NCHW (K4/K5/K10 of SURVEY
+// §2.4a): per-tap GEMM accumulation — the conv is 9 shifted GEMMs
+// O^T[cout][pix] += W^T_tap[cout][cin] x I[cin][pix + tap-shift] — so no
+// im2col materialization and no torch.cat/F.pad halo assembly (the
+// reference's concat+pad, /root/reference/distrifuser/modules/pp/conv2d.py:
+// 72-88, is pure overhead): halo rows are read IN PLACE from the comm
+// buffer via the top/bot pointers.
+//
+// Structure (mirrors the attention kernel's measured idioms,
+// profiles/attention_ladder_r01.md):
+// * SWAPPED operands: D[m=cout][n=pixel] = A[cout][cin] x B[cin][pixel]
+//   via mfma_f32_32x32x16_bf16, so output stores are x-contiguous
+//   (32 lanes = 64 B rows) and the WEIGHT operand is the register-friendly
+//   one: weights are prepacked host-side into exact per-lane fragment order
+//   ([tap][ks][ct][lane][8] — each fragment load is one coalesced 1 KB
+//   wave read, L2-resident across the grid).
+// * The input tile is staged once per cin-slice into LDS TRANSPOSED
+//   ([pixel-row][cin], b128-packed writes — the attention V^T staging
+//   pattern: global reads stay 128 B-coalesced along x, and each lane
+//   writes one b128 instead of 8 scalar ds_write_b16).
+// * B-fragment reads are row reads of the transposed tile; the XOR window
+//   swizzle spreads consecutive-row reads across the 16 B windows of each
+//   row (the 4-way floor of 128 B rows, same as the attention K tile).
+// * stride 2 splits the staged tile into x-parity PLANES so the
+//   stride-2 fragment reads become stride-1 row reads (no extra conflicts).
+// * Bias is deferred to the epilogue (fused add on the accumulators), which
+//   is also where per-channel SiLU could fuse later.
+//
+// Reference semantics matched: pp/conv2d.py:20-41 (sliced conv_in — same
+// kernel, interior pointer offset into the full latent), :59-112 (halo
+// exchange, zero-pad at true image borders).
+
+#include "common.h"
+#include "kernels.h"
+
+namespace {
+
+typedef float float16v __attribute__((ext_vector_type(16)));
+
+template <int ROW_BYTES>
+__device__ __forceinline__ int cv_swz(int row, int byte_off) {
+    static_assert((ROW_BYTES & (ROW_BYTES - 1)) == 0);
+    return byte_off ^ ((row & (ROW_BYTES / 16 - 1)) << 4);
+}
+
+// S: conv stride; YB x (XW*32) output tile per block; CIN_T cins per staged
+// LDS tile; NCT cout tiles (of 32) per block.
+template <int S, int YB, int XW, int CIN_T, int NCT>
+__global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Params p) {
+    constexpr int NW = YB * XW;
+    constexpr int XB = XW * 32;
+    constexpr int YIN = (YB - 1) * S + 3;          // staged input rows
+    constexpr int XIN = (XB - 1) * S + 3;          // staged input x positions
+    constexpr int XP = (S == 1) ? XIN : (XB + 1);  // rows per x-parity plane
+    constexpr int KS_T = CIN_T / 16;               // k-slices per staged tile
+    constexpr int ROW_B = CIN_T * 2;               // LDS row bytes
+    __shared__ char in_lds[S * YIN * XP * ROW_B];
+
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int lo = lane & 31;
+    const int hi = lane >> 5;
+    const int wy = wave / XW;
+    const int wx = wave % XW;
+
+    const int nxb = (p.Wo + XB - 1) / XB;
+    const int xb0 = (blockIdx.x % nxb) * XB;
+    const int yb0 = (blockIdx.x / nxb) * YB;
+    const int cb = blockIdx.y;
+    const int b = blockIdx.z;
+
+    const uint16_t* xin = p.x + (int64_t)b * p.x_sb;
+    const uint16_t* top = p.top ? p.top + (int64_t)b * p.t_sb : nullptr;
+    const uint16_t* bot = p.bot ? p.bot + (int64_t)b * p.b_sb : nullptr;
+
+    float16v acc[NCT] = {};
+
+    // weights are packed to KS 16-cin slices (KS*16 a multiple of 64, zero
+    // padded), so every staged tile is fully covered
+    const int n_cin_tiles = (p.KS * 16 + CIN_T - 1) / CIN_T;
+    for (int cint = 0; cint < n_cin_tiles; ++cint) {
+        const int cin0 = cint * CIN_T;
+        // ---- stage input tile [ry][xi][cin] transposed into LDS ----------
+        {
+            constexpr int CG = CIN_T / 8;  // 8-cin packs
+            constexpr int NITEMS = YIN * CG * XIN;
+            for (int c = tid; c < NITEMS; c += NW * WAVE_SIZE) {
+                const int xi = c % XIN;             // x fastest: coalesced
+                const int cg = (c / XIN) % CG;
+                const int ry = c / (XIN * CG);
+                const int y_in = yb0 * S - 1 + ry;
+                const int x_in = xb0 * S - 1 + xi;
+                const uint16_t* src = nullptr;
+                int64_t sc = 0;
+                int64_t row_off = 0;
+                if (x_in >= 0 && x_in < p.W) {
+                    if (y_in >= 0 && y_in < p.H) {
+                        src = xin;
+                        sc = p.x_sc;
+                        row_off = (int64_t)y_in * p.W + x_in;
+                    } else if (y_in == -1 && top) {
+                        src = top;
+                        sc = p.t_sc;
+                        row_off = x_in;
+                    } else if (y_in == p.H && bot) {
+                        src = bot;
+                        sc = p.b_sc;
+                        row_off = x_in;
+                    }
+                }
+                uint16_t vals[8];
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int cin = cin0 + cg * 8 + j;
+                    vals[j] = (src && cin < p.Cin)
+                                  ? src[(int64_t)cin * sc + row_off]
+                                  : (uint16_t)0;
+                }
+                const int plane = (S == 1) ? 0 : (xi & 1);
+                const int row = (S == 1) ? xi : (xi >> 1);
+                char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
+                                    cv_swz<ROW_B>(row, cg * 16)];
+                *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(vals);
+            }
+        }
+        __syncthreads();
+
+        // ---- accumulate 9 taps x KS_T k-slices ---------------------------
+#pragma unroll
+        for (int tap = 0; tap < 9; ++tap) {
+            const int dy = tap / 3;
+            const int dx = tap % 3;
+            const int ry = wy * S + dy;
+            const int xi = (wx * 32 + lo) * S + dx;
+            const int plane = (S == 1) ? 0 : (xi & 1);
+            const int row = (S == 1) ? xi : (xi >> 1);
+            const char* brow = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B];
+#pragma unroll
+            for (int ks2 = 0; ks2 < KS_T; ++ks2) {
+                const short8 bfrag = *reinterpret_cast<const short8*>(
+                    brow + cv_swz<ROW_B>(row, (ks2 * 16 + hi * 8) * 2));
+                const int ks = cint * KS_T + ks2;
+#pragma unroll
+                for (int ct2 = 0; ct2 < NCT; ++ct2) {
+                    const int ct = cb * NCT + ct2;
+                    if (ct < p.CT) {
+                        const short8 afrag = *reinterpret_cast<const short8*>(
+                            p.wp + (((int64_t)(tap * p.KS + ks) * p.CT + ct) * WAVE_SIZE + lane) * 8);
+                        acc[ct2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                            afrag, bfrag, acc[ct2], 0, 0, 0);
+                    }
+                }
+            }
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: O[b][cout][y][x] = acc + bias ---------------------------
+    const int y = yb0 + wy;
+    const int x = xb0 + wx * 32 + lo;
+    if (y >= p.Ho || x >= p.Wo) return;
+    uint16_t* obase = p.o + (int64_t)b * p.Cout * p.Ho * p.Wo + (int64_t)y * p.Wo + x;
+#pragma unroll
+    for (int ct2 = 0; ct2 < NCT; ++ct2) {
+        const int ct = cb * NCT + ct2;
+        if (ct >= p.CT) break;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            const int cout = ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            if (cout < p.Cout) {
+                float v = acc[ct2][r];
+                if (p.bias) v += to_f32(reinterpret_cast<const bf16_t*>(p.bias)[cout]);
+                obase[(int64_t)cout * p.Ho * p.Wo] =
+                    __builtin_bit_cast(uint16_t, __float2bfloat16(v));
+            }
+        }
+    }
+}
+
+}  // namespace
+
+void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
+    if (stride == 1) {
+        constexpr int YB = 2, XW = 4, NCT = 2;
+        const int nxb = (p.Wo + XW * 32 - 1) / (XW * 32);
+        const int nyb = (p.Ho + YB - 1) / YB;
+        dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
+        conv3x3_kernel<1, YB, XW, 64, NCT><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+    } else {
+        constexpr int YB = 2, XW = 4, NCT = 2;
+        const int nxb = (p.Wo + XW * 32 - 1) / (XW * 32);
+        const int nyb = (p.Ho + YB - 1) / YB;
+        dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
+        conv3x3_kernel<2, YB, XW, 32, NCT><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+    }
+}

@@ -41,6 +41,29 @@ void launch_cfg_affine_step(const void* noise_u, const void* noise_c, const void
                             float g, float ca, float cb, int64_t total, int dtype,
                             hipStream_t stream);
 
+// ---- Implicit-GEMM 3x3 conv (bf16, NCHW, stride 1/2, pad 1) -----------------
+// x: interior [B][Cin][H][W] (row-contiguous; x_sc = channel stride).
+// top/bot: optional single halo rows [B][Cin][1][W] (e.g. comm-buffer views);
+// null => zero padding at that border. Input row y_in=-1 reads top, y_in=H
+// reads bot. o: [B][Cout][Ho][Wo] contiguous output.
+// wp: weights prepacked in per-lane A-fragment order [9][KS][CT][64][8]
+// (cout = ct*32 + (lane&31), cin = ks*16 + (lane>>5)*8 + j), KS*16 and CT*32
+// zero-padded to multiples of 64/32. bias: bf16 [Cout] or null.
+struct Conv3x3Params {
+    const uint16_t* x;
+    const uint16_t* top;
+    const uint16_t* bot;
+    const uint16_t* wp;
+    const uint16_t* bias;
+    uint16_t* o;
+    int B, Cin, Cout, H, W, Ho, Wo;
+    int KS, CT;
+    int64_t x_sb, x_sc;
+    int64_t t_sb, t_sc;
+    int64_t b_sb, b_sc;
+};
+void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream);
+
 // ---- Flash attention (bf16, SD-family head dims) ----------------------------
 // q: logical [B, H, Lq, 64]; k/v: logical [B, H, NC, LC, 64] (NC stale-KV
 // chunks of LC tokens; NC=1 for plain attention). All strides in ELEMENTS,

@@ -16,6 +16,7 @@ import torch
 from torch import nn
 
 from .. import ops
+from ..ops import NativeConv2d
 from ..parallel.patch_ops import (
     CachedCrossAttention,
     PatchConv2d,
@@ -126,7 +127,7 @@ class LayerFactory:
             and in_ch % self.state.config.n_device_per_batch == 0
         ):
             return TPConv2d(in_ch, out_ch, kernel, stride, padding, bias, state=self.state)
-        return nn.Conv2d(in_ch, out_ch, kernel, stride=stride, padding=padding, bias=bias)
+        return NativeConv2d(in_ch, out_ch, kernel, stride=stride, padding=padding, bias=bias)
 
     def group_norm(self, groups: int, ch: int, eps: float = 1e-5, fuse_silu: bool = False):
         if self.parallelism == "patch":

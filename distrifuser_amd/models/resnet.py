@@ -13,6 +13,7 @@ import torch.distributed as dist
 import torch.nn.functional as F
 from torch import nn
 
+from ..ops import NativeConv2d
 from .layers import LayerFactory
 
 
@@ -53,10 +54,10 @@ class ResnetBlock2D(nn.Module):
             rank = state.config.split_idx()
             self.c_local = out_channels // n
             self.c_start = rank * self.c_local
-            self.conv1 = nn.Conv2d(in_channels, self.c_local, 3, padding=1, bias=True)
+            self.conv1 = NativeConv2d(in_channels, self.c_local, 3, padding=1, bias=True)
             self.time_emb_proj = nn.Linear(temb_channels, self.c_local)
             self.norm2 = nn.GroupNorm(groups // n, self.c_local, eps=eps)
-            self.conv2 = nn.Conv2d(self.c_local, out_channels, 3, padding=1, bias=False)
+            self.conv2 = NativeConv2d(self.c_local, out_channels, 3, padding=1, bias=False)
             self.conv2_bias = nn.Parameter(torch.zeros(out_channels))
         else:
             self.conv1 = factory.conv2d(in_channels, out_channels, 3, 1, 1)
@@ -64,7 +65,7 @@ class ResnetBlock2D(nn.Module):
             self.norm2 = factory.group_norm(groups, out_channels, eps=eps, fuse_silu=True)
             self.conv2 = factory.conv2d(out_channels, out_channels, 3, 1, 1)
         if in_channels != out_channels:
-            self.conv_shortcut = nn.Conv2d(in_channels, out_channels, 1)
+            self.conv_shortcut = NativeConv2d(in_channels, out_channels, 1)
         else:
             self.conv_shortcut = None
 

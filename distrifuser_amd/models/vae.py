@@ -16,6 +16,8 @@ import torch
 import torch.nn.functional as F
 from torch import nn
 
+from ..ops import NativeConv2d
+
 from .. import ops
 
 
@@ -44,10 +46,10 @@ class VAEResnetBlock(nn.Module):
     def __init__(self, in_ch: int, out_ch: int, groups: int):
         super().__init__()
         self.norm1 = nn.GroupNorm(groups, in_ch, eps=1e-6)
-        self.conv1 = nn.Conv2d(in_ch, out_ch, 3, padding=1)
+        self.conv1 = NativeConv2d(in_ch, out_ch, 3, padding=1)
         self.norm2 = nn.GroupNorm(groups, out_ch, eps=1e-6)
-        self.conv2 = nn.Conv2d(out_ch, out_ch, 3, padding=1)
-        self.conv_shortcut = nn.Conv2d(in_ch, out_ch, 1) if in_ch != out_ch else None
+        self.conv2 = NativeConv2d(out_ch, out_ch, 3, padding=1)
+        self.conv_shortcut = NativeConv2d(in_ch, out_ch, 1) if in_ch != out_ch else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         h = self.conv1(ops.group_norm_silu(x, self.norm1.num_groups, self.norm1.weight,
@@ -116,7 +118,7 @@ class UpDecoderBlock2D(nn.Module):
         self.resnets = nn.ModuleList(
             [VAEResnetBlock(in_ch if i == 0 else out_ch, out_ch, groups) for i in range(layers)]
         )
-        self.upsampler = nn.Conv2d(out_ch, out_ch, 3, padding=1) if add_upsample else None
+        self.upsampler = NativeConv2d(out_ch, out_ch, 3, padding=1) if add_upsample else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         for r in self.resnets:
@@ -134,8 +136,8 @@ class VAEDecoder(nn.Module):
         ch = config.block_out_channels
         groups = config.norm_num_groups
         top = ch[-1]
-        self.post_quant_conv = nn.Conv2d(config.latent_channels, config.latent_channels, 1)
-        self.conv_in = nn.Conv2d(config.latent_channels, top, 3, padding=1)
+        self.post_quant_conv = NativeConv2d(config.latent_channels, config.latent_channels, 1)
+        self.conv_in = NativeConv2d(config.latent_channels, top, 3, padding=1)
         self.mid_resnet_1 = VAEResnetBlock(top, top, groups)
         self.mid_attn = VAEAttention(top, groups)
         self.mid_resnet_2 = VAEResnetBlock(top, top, groups)
@@ -152,7 +154,7 @@ class VAEDecoder(nn.Module):
             prev = out_ch
         self.up_blocks = nn.ModuleList(blocks)
         self.conv_norm_out = nn.GroupNorm(groups, ch[0], eps=1e-6)
-        self.conv_out = nn.Conv2d(ch[0], config.out_channels, 3, padding=1)
+        self.conv_out = NativeConv2d(ch[0], config.out_channels, 3, padding=1)
 
     @torch.no_grad()
     def decode(self, latents: torch.Tensor) -> torch.Tensor:

@@ -7,6 +7,7 @@ runs, which is what the non-GPU test suite checks numerics against.
 Set DFA_FORCE_EAGER=1 to force the eager path on GPU (A/B debugging only).
 """
 
+from .conv import NativeConv2d, conv3x3_halo, pack_conv3x3_weight
 from .dispatch import (
     flash_attention,
     flash_attention_chunked,
@@ -19,6 +20,8 @@ from .dispatch import (
 )
 
 __all__ = [
+    "NativeConv2d",
+    "conv3x3_halo",
     "flash_attention",
     "flash_attention_chunked",
     "geglu",
@@ -27,4 +30,5 @@ __all__ = [
     "group_norm_stats",
     "hip_ext",
     "hip_ext_available",
+    "pack_conv3x3_weight",
 ]

@@ -87,3 +87,36 @@ def geglu(hidden: torch.Tensor) -> torch.Tensor:
     """GEGLU gate: split last dim in half, return a * gelu(b) (tanh=false)."""
     a, b = hidden.chunk(2, dim=-1)
     return a * F.gelu(b)
+
+
+def conv3x3_halo(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: torch.Tensor | None,
+    stride: int = 1,
+    top: torch.Tensor | None = None,
+    bot: torch.Tensor | None = None,
+) -> torch.Tensor:
+    """3x3 pad-1 conv whose top/bottom halo rows come from separate tensors.
+
+    x: [B, Cin, H, W]; top/bot: [B, Cin, 1, W] neighbour rows (None => zero
+    pad at that border). Semantics of the HIP conv3x3 kernel: the reference
+    materializes cat([top, x, bot]) per conv
+    (/root/reference/distrifuser/modules/pp/conv2d.py:72-88); the kernel —
+    and this oracle — treat the halos as extra input rows in place.
+    """
+    b, c, h, w = x.shape
+    parts = []
+    pad_top = 1 if top is None else 0
+    pad_bot = 1 if bot is None else 0
+    if top is not None:
+        parts.append(top.reshape(b, c, 1, w))
+    parts.append(x)
+    if bot is not None:
+        parts.append(bot.reshape(b, c, 1, w))
+    full = torch.cat(parts, dim=2) if len(parts) > 1 else x
+    full = F.pad(full, [1, 1, pad_top, pad_bot])
+    out = F.conv2d(full, weight, bias, stride=stride)
+    # pad-1 stride-s output has ceil(H/s) rows regardless of halo presence
+    ho = (h - 1) // stride + 1
+    return out[:, :, :ho, :]

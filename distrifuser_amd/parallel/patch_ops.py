@@ -62,7 +62,7 @@ class PatchConv2d(nn.Module):
         is_first_layer: bool = False,
     ):
         super().__init__()
-        self.conv = nn.Conv2d(
+        self.conv = ops.NativeConv2d(
             in_channels, out_channels, kernel_size, stride=stride, padding=padding, bias=bias
         )
         self.state = state
@@ -161,16 +161,9 @@ class PatchConv2d(nn.Module):
             padded = F.pad(padded, [0, 0, pad_top, pad_bot])
             return F.conv2d(padded, conv.weight, conv.bias, stride=s, padding=(0, pw))
 
-        out = F.conv2d(x, conv.weight, conv.bias, stride=s, padding=(1, pw))
-        # first output row: window covers rows [-1, 1] -> needs the top halo
-        if top is not None:
-            tin = torch.cat([top, x[:, :, :2]], dim=2)
-            out[:, :, :1] = F.conv2d(tin, conv.weight, conv.bias, stride=s, padding=(0, pw))
-        # last output row (stride 1 only; stride 2's last window ends inside x)
-        if s == 1 and bot is not None:
-            bin_ = torch.cat([x[:, :, -2:], bot], dim=2)
-            out[:, :, -1:] = F.conv2d(bin_, conv.weight, conv.bias, stride=s, padding=(0, pw))
-        return out
+        # NativeConv2d reads the halo rows in place (HIP kernel: the top/bot
+        # pointers, SURVEY K4 — no cat, no boundary-row recompute pass)
+        return conv(x, top=top, bot=bot)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if not _is_patch_parallel(self.state):

@@ -20,6 +20,7 @@ import torch.distributed as dist
 import torch.nn.functional as F
 from torch import nn
 
+from .. import ops
 from .state import ParallelState
 
 
@@ -178,7 +179,7 @@ class TPConv2d(nn.Module):
         assert in_channels % n == 0, f"conv in_channels {in_channels} must divide TP degree {n}"
         self.c_local = in_channels // n
         self.c_start = rank * self.c_local
-        self.conv = nn.Conv2d(
+        self.conv = ops.NativeConv2d(
             self.c_local, out_channels, kernel_size, stride=stride, padding=padding, bias=False
         )
         self.bias = nn.Parameter(torch.zeros(out_channels)) if bias else None

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "geglu.hip"),
         os.path.join(CSRC, "scheduler.hip"),
         os.path.join(CSRC, "attention.hip"),
+        os.path.join(CSRC, "conv.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++20"],

@@ -65,3 +65,28 @@ def test_group_norm_stats_distributed_average_equals_full():
     bot = eager.group_norm_stats(x[:, :, 4:], g)
     avg = (top + bot) / 2
     assert torch.allclose(avg, full, atol=1e-6)
+
+
+def test_conv3x3_halo_oracle_matches_full_conv():
+    """conv3x3_halo with neighbour halo rows == conv over the full tensor."""
+    import torch.nn.functional as F
+
+    from distrifuser_amd.ops import eager
+
+    torch.manual_seed(0)
+    full = torch.randn(2, 16, 12, 20)
+    w = torch.randn(24, 16, 3, 3) * 0.1
+    b = torch.randn(24)
+    for stride in (1, 2):
+        ref = F.conv2d(full, w, b, stride=stride, padding=1)
+        # split rows 4..8 as the "local band"; halos from the neighbours
+        x = full[:, :, 4:8]
+        top = full[:, :, 3:4]
+        bot = full[:, :, 8:9]
+        got = eager.conv3x3_halo(x, w, b, stride, top, bot if stride == 1 else None)
+        band = ref[:, :, 4 // stride : 8 // stride]
+        assert torch.allclose(got, band, atol=1e-4), (stride, (got - band).abs().max())
+    # border band: no top halo -> zero padding must match the full conv edge
+    x0 = full[:, :, :4]
+    got0 = eager.conv3x3_halo(x0, w, b, 1, None, full[:, :, 4:5])
+    assert torch.allclose(got0, F.conv2d(full, w, b, padding=1)[:, :, :4], atol=1e-4)
