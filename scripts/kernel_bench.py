@@ -77,6 +77,34 @@ def main():
         })
         print(json.dumps(results[-1]), flush=True)
 
+    # ---- implicit-GEMM conv3x3 at SDXL 3840^2 shapes (2-sample CFG batch) ----
+    torch.backends.cudnn.benchmark = True
+    conv_shapes = [
+        (4, 320, 480, 480, 1),      # conv_in
+        (320, 320, 480, 480, 1),    # down0 ResBlock
+        (320, 320, 240, 240, 2),    # downsample
+        (640, 640, 240, 240, 1),
+        (1280, 1280, 120, 120, 1),  # mid/up ResBlocks
+        (2560, 1280, 120, 120, 1),  # up-block concat conv
+        (960, 320, 480, 480, 1),
+    ]
+    for cin, cout, h, w, s in conv_shapes:
+        x = torch.randn(2, cin, h, w, device=dev, dtype=torch.bfloat16) * 0.5
+        wt = torch.randn(cout, cin, 3, 3, device=dev, dtype=torch.bfloat16) * (cin * 9) ** -0.5
+        bias = torch.randn(cout, device=dev, dtype=torch.bfloat16)
+        packed = ops.pack_conv3x3_weight(wt)
+        ms_ours = timeit(lambda: ops.conv3x3_halo(x, wt, bias, s, packed=packed))
+        ms_ref = timeit(lambda: F.conv2d(x, wt, bias, stride=s, padding=1))
+        ho, wo = (h - 1) // s + 1, (w - 1) // s + 1
+        flops = 2.0 * 2 * cout * ho * wo * cin * 9
+        results.append({
+            "op": "conv3x3", "cin": cin, "cout": cout, "h": h, "w": w, "stride": s,
+            "ms_ours": ms_ours, "ms_miopen": ms_ref,
+            "tflops_ours": flops / ms_ours / 1e9,
+            "tflops_miopen": flops / ms_ref / 1e9,
+        })
+        print(json.dumps(results[-1]), flush=True)
+
     # ---- fused GN+SiLU at SDXL shapes ----
     for c, hw in [(320, 480), (640, 240), (1280, 120), (320, 128), (640, 64)]:
         x = torch.randn(2, c, hw, hw, device=dev, dtype=torch.bfloat16)
