@@ -43,6 +43,62 @@ __device__ __forceinline__ int cv_swz(int row, int byte_off) {
     return byte_off ^ ((row & (ROW_BYTES / 16 - 1)) << 4);
 }
 
+// In-register 8x8 bf16 transpose across 8 consecutive lanes (butterfly):
+// lane i of the group holds row i (8 elements = one uint4); afterwards lane
+// i holds column i. Three stages (element/lane bit b in {1,2,4}); per stage
+// where bit(lane,b) != bit(elem,b): new[c] = partner(lane^b)'s e[c^b] —
+// implemented as local bit-b element swap, shfl_xor(b), select off slots.
+// Replaces the 8 scalar 2-byte global loads per staged item of the v1
+// staging (the measured bottleneck: 122 TF with scalar-gather staging).
+__device__ __forceinline__ uint4 transpose8x8_bf16(uint4 v, int lane) {
+    uint32_t d[4] = {v.x, v.y, v.z, v.w};
+    // stage b=1: elements swap within dwords (bf16 halves)
+    {
+        const bool hi = (lane & 1) != 0;
+        uint32_t recv[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const uint32_t sw = (d[i] >> 16) | (d[i] << 16);  // e[c^1]
+            recv[i] = __shfl_xor(sw, 1, WAVE_SIZE);
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+            d[i] = hi ? (d[i] & 0xffff0000u) | (recv[i] & 0x0000ffffu)
+                      : (d[i] & 0x0000ffffu) | (recv[i] & 0xffff0000u);
+    }
+    // stage b=2: dword pairs (0<->1, 2<->3)
+    {
+        const bool hi = (lane & 2) != 0;
+        const uint32_t r0 = __shfl_xor(d[1], 2, WAVE_SIZE);  // partner e[0^2]
+        const uint32_t r1 = __shfl_xor(d[0], 2, WAVE_SIZE);
+        const uint32_t r2 = __shfl_xor(d[3], 2, WAVE_SIZE);
+        const uint32_t r3 = __shfl_xor(d[2], 2, WAVE_SIZE);
+        if (hi) {
+            d[0] = r0;
+            d[2] = r2;
+        } else {
+            d[1] = r1;
+            d[3] = r3;
+        }
+    }
+    // stage b=4: dword pairs (0<->2, 1<->3)
+    {
+        const bool hi = (lane & 4) != 0;
+        const uint32_t r0 = __shfl_xor(d[2], 4, WAVE_SIZE);
+        const uint32_t r1 = __shfl_xor(d[3], 4, WAVE_SIZE);
+        const uint32_t r2 = __shfl_xor(d[0], 4, WAVE_SIZE);
+        const uint32_t r3 = __shfl_xor(d[1], 4, WAVE_SIZE);
+        if (hi) {
+            d[0] = r0;
+            d[1] = r1;
+        } else {
+            d[2] = r2;
+            d[3] = r3;
+        }
+    }
+    return uint4{d[0], d[1], d[2], d[3]};
+}
+
 // S: conv stride; YB x (XW*32) output tile per block; CIN_T cins per staged
 // LDS tile; NCT cout tiles (of 32) per block.
 template <int S, int YB, int XW, int CIN_T, int NCT>
@@ -82,46 +138,83 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     for (int cint = 0; cint < n_cin_tiles; ++cint) {
         const int cin0 = cint * CIN_T;
         // ---- stage input tile [ry][xi][cin] transposed into LDS ----------
+        // Vectorized: each lane b128-loads 8 consecutive x of one cin row
+        // (coalesced 128 B segments), an in-register 8x8 butterfly flips the
+        // slab to [x][cin], one b128 LDS write per lane. Edges / W%8!=0 take
+        // guarded scalar element loads inside the same slab.
         {
-            constexpr int CG = CIN_T / 8;  // 8-cin packs
-            constexpr int NITEMS = YIN * CG * XIN;
-            for (int c = tid; c < NITEMS; c += NW * WAVE_SIZE) {
-                const int xi = c % XIN;             // x fastest: coalesced
-                const int cg = (c / XIN) % CG;
-                const int ry = c / (XIN * CG);
+            constexpr int CG = CIN_T / 8;            // 8-cin groups
+            constexpr int NXW = (XIN - 1 + 7) / 8;   // 8-x windows for xi>=1
+            constexpr int NXS = (NXW + 7) / 8;       // windows per wave pass
+            const bool w_vec = (p.W % 8) == 0;
+            for (int slab = wave; slab < YIN * CG * NXS; slab += NW) {
+                const int xs = slab % NXS;
+                const int cg = (slab / NXS) % CG;
+                const int ry = slab / (NXS * CG);
                 const int y_in = yb0 * S - 1 + ry;
-                const int x_in = xb0 * S - 1 + xi;
                 const uint16_t* src = nullptr;
                 int64_t sc = 0;
-                int64_t row_off = 0;
-                if (x_in >= 0 && x_in < p.W) {
-                    if (y_in >= 0 && y_in < p.H) {
-                        src = xin;
-                        sc = p.x_sc;
-                        row_off = (int64_t)y_in * p.W + x_in;
-                    } else if (y_in == -1 && top) {
-                        src = top;
-                        sc = p.t_sc;
-                        row_off = x_in;
-                    } else if (y_in == p.H && bot) {
-                        src = bot;
-                        sc = p.b_sc;
-                        row_off = x_in;
+                int64_t rbase = 0;
+                if (y_in >= 0 && y_in < p.H) {
+                    src = xin;
+                    sc = p.x_sc;
+                    rbase = (int64_t)y_in * p.W;
+                } else if (y_in == -1 && top) {
+                    src = top;
+                    sc = p.t_sc;
+                } else if (y_in == p.H && bot) {
+                    src = bot;
+                    sc = p.b_sc;
+                }
+                const int xw = xs * 8 + (lane >> 3);     // this lane's window
+                const int cin = cin0 + cg * 8 + (lane & 7);
+                const int x_in0 = xb0 * S + xw * 8;      // xi = 1 + 8*xw
+                uint4 raw = {0, 0, 0, 0};
+                if (src && cin < p.Cin && 8 * xw + 1 < XIN) {
+                    const uint16_t* rp = src + (int64_t)cin * sc + rbase;
+                    if (w_vec && x_in0 + 8 <= p.W) {
+                        raw = *reinterpret_cast<const uint4*>(rp + x_in0);
+                    } else {
+                        uint16_t vals[8];
+#pragma unroll
+                        for (int j = 0; j < 8; ++j) {
+                            const int x_in = x_in0 + j;
+                            vals[j] = (x_in < p.W) ? rp[x_in] : (uint16_t)0;
+                        }
+                        raw = *reinterpret_cast<const uint4*>(vals);
                     }
                 }
-                uint16_t vals[8];
-#pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int cin = cin0 + cg * 8 + j;
-                    vals[j] = (src && cin < p.Cin)
-                                  ? src[(int64_t)cin * sc + row_off]
-                                  : (uint16_t)0;
+                const uint4 tr = transpose8x8_bf16(raw, lane);
+                const int xi = 1 + xw * 8 + (lane & 7);  // lane now owns x=xi
+                if (xi < XIN) {
+                    const int plane = (S == 1) ? 0 : (xi & 1);
+                    const int row = (S == 1) ? xi : (xi >> 1);
+                    char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
+                                        cv_swz<ROW_B>(row, cg * 16)];
+                    *reinterpret_cast<uint4*>(dst) = tr;
                 }
-                const int plane = (S == 1) ? 0 : (xi & 1);
-                const int row = (S == 1) ? xi : (xi >> 1);
+            }
+            // left halo column xi = 0 (x_in = xb0*S - 1): scalar, tiny
+            for (int c = tid; c < YIN * CIN_T; c += NW * WAVE_SIZE) {
+                const int ci = c % CIN_T;
+                const int ry = c / CIN_T;
+                const int y_in = yb0 * S - 1 + ry;
+                const int x_in = xb0 * S - 1;
+                const int cin = cin0 + ci;
+                uint16_t val = 0;
+                if (x_in >= 0 && cin < p.Cin) {
+                    if (y_in >= 0 && y_in < p.H)
+                        val = xin[(int64_t)cin * p.x_sc + (int64_t)y_in * p.W + x_in];
+                    else if (y_in == -1 && top)
+                        val = top[(int64_t)cin * p.t_sc + x_in];
+                    else if (y_in == p.H && bot)
+                        val = bot[(int64_t)cin * p.b_sc + x_in];
+                }
+                const int plane = (S == 1) ? 0 : 0;  // xi=0 is even
+                const int row = 0;
                 char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
-                                    cv_swz<ROW_B>(row, cg * 16)];
-                *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(vals);
+                                    cv_swz<ROW_B>(row, (ci / 8) * 16) + (ci % 8) * 2];
+                *reinterpret_cast<uint16_t*>(dst) = val;
             }
         }
         __syncthreads();
