@@ -99,18 +99,26 @@ __device__ __forceinline__ uint4 transpose8x8_bf16(uint4 v, int lane) {
     return uint4{d[0], d[1], d[2], d[3]};
 }
 
-// S: conv stride; YB x (XW*32) output tile per block; CIN_T cins per staged
-// LDS tile; NCT cout tiles (of 32) per block.
-template <int S, int YB, int XW, int CIN_T, int NCT>
+// S: conv stride; (YB rows) x (XW*32*PW x) output tile per block; CIN_T cins
+// per staged LDS tile; each wave owns PW 32-x subtiles x NCT 32-cout tiles
+// (A fragments reused across the PW subtiles, B across the NCT tiles).
+// Weight fragments for the (cin-tile, cout-block) are staged into LDS
+// cooperatively alongside the input tile: the v2 kernel loaded each A
+// fragment from L2 right before its MFMA and stalled on latency (PMC:
+// MFMA util 6%, SQ_WAIT dominated — see profiles/ notes); from LDS the
+// fragment read is a contiguous wave-wide 1 KB burst.
+template <int S, int YB, int XW, int CIN_T, int NCT, int PW>
 __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Params p) {
     constexpr int NW = YB * XW;
-    constexpr int XB = XW * 32;
+    constexpr int XB = XW * 32 * PW;
     constexpr int YIN = (YB - 1) * S + 3;          // staged input rows
     constexpr int XIN = (XB - 1) * S + 3;          // staged input x positions
     constexpr int XP = (S == 1) ? XIN : (XB + 1);  // rows per x-parity plane
     constexpr int KS_T = CIN_T / 16;               // k-slices per staged tile
     constexpr int ROW_B = CIN_T * 2;               // LDS row bytes
+    constexpr int NFRAG = 9 * KS_T * NCT;          // weight frags per tile
     __shared__ char in_lds[S * YIN * XP * ROW_B];
+    __shared__ char w_lds[NFRAG * WAVE_SIZE * 16];
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
@@ -130,17 +138,36 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     const uint16_t* top = p.top ? p.top + (int64_t)b * p.t_sb : nullptr;
     const uint16_t* bot = p.bot ? p.bot + (int64_t)b * p.b_sb : nullptr;
 
-    float16v acc[NCT] = {};
+    float16v acc[NCT][PW] = {};
 
     // weights are packed to KS 16-cin slices (KS*16 a multiple of 64, zero
     // padded), so every staged tile is fully covered
     const int n_cin_tiles = (p.KS * 16 + CIN_T - 1) / CIN_T;
     for (int cint = 0; cint < n_cin_tiles; ++cint) {
         const int cin0 = cint * CIN_T;
+        // ---- stage weight fragments [tap][ks2][ct2][lane] into LDS -------
+        {
+            for (int t = tid; t < NFRAG * WAVE_SIZE; t += NW * WAVE_SIZE) {
+                const int frag = t >> 6;
+                const int l = t & 63;
+                const int ct2 = frag % NCT;
+                const int ks2 = (frag / NCT) % KS_T;
+                const int tap = frag / (NCT * KS_T);
+                const int ct = cb * NCT + ct2;
+                uint4 w4 = {0, 0, 0, 0};
+                if (ct < p.CT) {
+                    const int64_t gidx =
+                        ((int64_t)(tap * p.KS + cint * KS_T + ks2) * p.CT + ct) * WAVE_SIZE + l;
+                    w4 = *reinterpret_cast<const uint4*>(p.wp + gidx * 8);
+                }
+                // LDS layout mirrors the compute loop: [tap][ks2][ct2][l]
+                *reinterpret_cast<uint4*>(&w_lds[(int64_t)t * 16]) = w4;
+            }
+        }
         // ---- stage input tile [ry][xi][cin] transposed into LDS ----------
-        // Vectorized: each lane b128-loads 8 consecutive x of one cin row
-        // (coalesced 128 B segments), an in-register 8x8 butterfly flips the
-        // slab to [x][cin], one b128 LDS write per lane. Edges / W%8!=0 take
+        // Each lane b128-loads 8 consecutive x of one cin row (coalesced
+        // 128 B segments), an in-register 8x8 butterfly flips the slab to
+        // [x][cin], one b128 LDS write per lane. Edges / W%8!=0 take
         // guarded scalar element loads inside the same slab.
         {
             constexpr int CG = CIN_T / 8;            // 8-cin groups
@@ -210,9 +237,8 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                     else if (y_in == p.H && bot)
                         val = bot[(int64_t)cin * p.b_sc + x_in];
                 }
-                const int plane = (S == 1) ? 0 : 0;  // xi=0 is even
-                const int row = 0;
-                char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
+                const int row = 0;  // xi = 0: plane 0, row 0 for both strides
+                char* dst = &in_lds[((0 * YIN + ry) * XP + row) * ROW_B +
                                     cv_swz<ROW_B>(row, (ci / 8) * 16) + (ci % 8) * 2];
                 *reinterpret_cast<uint16_t*>(dst) = val;
             }
@@ -225,24 +251,26 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
             const int dy = tap / 3;
             const int dx = tap % 3;
             const int ry = wy * S + dy;
-            const int xi = (wx * 32 + lo) * S + dx;
-            const int plane = (S == 1) ? 0 : (xi & 1);
-            const int row = (S == 1) ? xi : (xi >> 1);
-            const char* brow = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B];
 #pragma unroll
             for (int ks2 = 0; ks2 < KS_T; ++ks2) {
-                const short8 bfrag = *reinterpret_cast<const short8*>(
-                    brow + cv_swz<ROW_B>(row, (ks2 * 16 + hi * 8) * 2));
-                const int ks = cint * KS_T + ks2;
+                short8 bfrag[PW];
+#pragma unroll
+                for (int px = 0; px < PW; ++px) {
+                    const int xi = ((wx * PW + px) * 32 + lo) * S + dx;
+                    const int plane = (S == 1) ? 0 : (xi & 1);
+                    const int row = (S == 1) ? xi : (xi >> 1);
+                    bfrag[px] = *reinterpret_cast<const short8*>(
+                        &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
+                                cv_swz<ROW_B>(row, (ks2 * 16 + hi * 8) * 2)]);
+                }
 #pragma unroll
                 for (int ct2 = 0; ct2 < NCT; ++ct2) {
-                    const int ct = cb * NCT + ct2;
-                    if (ct < p.CT) {
-                        const short8 afrag = *reinterpret_cast<const short8*>(
-                            p.wp + (((int64_t)(tap * p.KS + ks) * p.CT + ct) * WAVE_SIZE + lane) * 8);
-                        acc[ct2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                            afrag, bfrag, acc[ct2], 0, 0, 0);
-                    }
+                    const short8 afrag = *reinterpret_cast<const short8*>(
+                        &w_lds[(((tap * KS_T + ks2) * NCT + ct2) * WAVE_SIZE + lane) * 16]);
+#pragma unroll
+                    for (int px = 0; px < PW; ++px)
+                        acc[ct2][px] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                            afrag, bfrag[px], acc[ct2][px], 0, 0, 0);
                 }
             }
         }
@@ -251,21 +279,25 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
 
     // ---- epilogue: O[b][cout][y][x] = acc + bias ---------------------------
     const int y = yb0 + wy;
-    const int x = xb0 + wx * 32 + lo;
-    if (y >= p.Ho || x >= p.Wo) return;
-    uint16_t* obase = p.o + (int64_t)b * p.Cout * p.Ho * p.Wo + (int64_t)y * p.Wo + x;
+    if (y >= p.Ho) return;
 #pragma unroll
-    for (int ct2 = 0; ct2 < NCT; ++ct2) {
-        const int ct = cb * NCT + ct2;
-        if (ct >= p.CT) break;
+    for (int px = 0; px < PW; ++px) {
+        const int x = xb0 + (wx * PW + px) * 32 + lo;
+        if (x >= p.Wo) continue;
+        uint16_t* obase = p.o + (int64_t)b * p.Cout * p.Ho * p.Wo + (int64_t)y * p.Wo + x;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-            const int cout = ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-            if (cout < p.Cout) {
-                float v = acc[ct2][r];
-                if (p.bias) v += to_f32(reinterpret_cast<const bf16_t*>(p.bias)[cout]);
-                obase[(int64_t)cout * p.Ho * p.Wo] =
-                    __builtin_bit_cast(uint16_t, __float2bfloat16(v));
+        for (int ct2 = 0; ct2 < NCT; ++ct2) {
+            const int ct = cb * NCT + ct2;
+            if (ct >= p.CT) break;
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                const int cout = ct * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                if (cout < p.Cout) {
+                    float v = acc[ct2][px][r];
+                    if (p.bias) v += to_f32(reinterpret_cast<const bf16_t*>(p.bias)[cout]);
+                    obase[(int64_t)cout * p.Ho * p.Wo] =
+                        __builtin_bit_cast(uint16_t, __float2bfloat16(v));
+                }
             }
         }
     }
@@ -275,16 +307,22 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
 
 void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
     if (stride == 1) {
-        constexpr int YB = 2, XW = 4, NCT = 2;
-        const int nxb = (p.Wo + XW * 32 - 1) / (XW * 32);
+        // 8 waves: 8 output rows x 1 x-wave of 64 px (PW=2 A-frag reuse);
+        // LDS: input 10x66x128 B = 84.5 KB + 72 KB weight frags = 156.5 KB.
+        constexpr int YB = 8, XW = 1, NCT = 2, PW = 2;
+        constexpr int XB = XW * 32 * PW;
+        const int nxb = (p.Wo + XB - 1) / XB;
         const int nyb = (p.Ho + YB - 1) / YB;
         dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
-        conv3x3_kernel<1, YB, XW, 64, NCT><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+        conv3x3_kernel<1, YB, XW, 64, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     } else {
-        constexpr int YB = 2, XW = 4, NCT = 2;
-        const int nxb = (p.Wo + XW * 32 - 1) / (XW * 32);
+        // stride 2: 8 waves: 4 output rows x 2 x-waves of 32 px;
+        // LDS: parity-plane input 74.9 KB + 36 KB weight frags.
+        constexpr int YB = 4, XW = 2, NCT = 2, PW = 1;
+        constexpr int XB = XW * 32 * PW;
+        const int nxb = (p.Wo + XB - 1) / XB;
         const int nyb = (p.Ho + YB - 1) / YB;
         dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
-        conv3x3_kernel<2, YB, XW, 32, NCT><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+        conv3x3_kernel<2, YB, XW, 32, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     }
 }
