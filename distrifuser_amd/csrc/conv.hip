@@ -43,36 +43,45 @@ __device__ __forceinline__ int cv_swz(int row, int byte_off) {
     return byte_off ^ ((row & (ROW_BYTES / 16 - 1)) << 4);
 }
 
-// In-register 8x8 bf16 transpose across 8 consecutive lanes (butterfly):
-// lane i of the group holds row i (8 elements = one uint4); afterwards lane
-// i holds column i. Three stages (element/lane bit b in {1,2,4}); per stage
-// where bit(lane,b) != bit(elem,b): new[c] = partner(lane^b)'s e[c^b] —
-// implemented as local bit-b element swap, shfl_xor(b), select off slots.
-// Replaces the 8 scalar 2-byte global loads per staged item of the v1
-// staging (the measured bottleneck: 122 TF with scalar-gather staging).
+// In-register 8x8 bf16 transpose across an 8-lane group (butterfly): group
+// member r = (lane&3)|((lane&8)>>1) holds row r (8 elements = one uint4);
+// afterwards member r holds column r. Lane-mask set {1,2,8} is chosen so
+// every exchange is a VALU DPP op (quad_perm for ^1/^2, row_ror:8 for ^8)
+// — __shfl_xor compiles to ds_bpermute, which rides the SAME LDS pipe the
+// MFMA fragment reads need (v3 PMC: 2.9 LDS instrs/MFMA, 19% MFMA util).
+// Stage (lane-bit m, elem-bit e): where bit(lane,m) != bit(elem,e):
+// new[c] = partner(lane^m)'s e[c^e]; composition over the three stages is
+// the full transpose (verified by simulation).
+__device__ __forceinline__ uint32_t dpp_xor1(uint32_t v) {
+    return __builtin_amdgcn_mov_dpp(v, 0xB1, 0xf, 0xf, true);  // quad_perm [1,0,3,2]
+}
+__device__ __forceinline__ uint32_t dpp_xor2(uint32_t v) {
+    return __builtin_amdgcn_mov_dpp(v, 0x4E, 0xf, 0xf, true);  // quad_perm [2,3,0,1]
+}
+__device__ __forceinline__ uint32_t dpp_xor8(uint32_t v) {
+    return __builtin_amdgcn_mov_dpp(v, 0x128, 0xf, 0xf, true);  // row_ror:8
+}
+
 __device__ __forceinline__ uint4 transpose8x8_bf16(uint4 v, int lane) {
     uint32_t d[4] = {v.x, v.y, v.z, v.w};
-    // stage b=1: elements swap within dwords (bf16 halves)
+    // stage (m=1, e=1): bf16 halves within dwords
     {
         const bool hi = (lane & 1) != 0;
-        uint32_t recv[4];
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
             const uint32_t sw = (d[i] >> 16) | (d[i] << 16);  // e[c^1]
-            recv[i] = __shfl_xor(sw, 1, WAVE_SIZE);
+            const uint32_t recv = dpp_xor1(sw);
+            d[i] = hi ? (d[i] & 0xffff0000u) | (recv & 0x0000ffffu)
+                      : (d[i] & 0x0000ffffu) | (recv & 0xffff0000u);
         }
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-            d[i] = hi ? (d[i] & 0xffff0000u) | (recv[i] & 0x0000ffffu)
-                      : (d[i] & 0x0000ffffu) | (recv[i] & 0xffff0000u);
     }
-    // stage b=2: dword pairs (0<->1, 2<->3)
+    // stage (m=2, e=2): dword pairs (0<->1, 2<->3)
     {
         const bool hi = (lane & 2) != 0;
-        const uint32_t r0 = __shfl_xor(d[1], 2, WAVE_SIZE);  // partner e[0^2]
-        const uint32_t r1 = __shfl_xor(d[0], 2, WAVE_SIZE);
-        const uint32_t r2 = __shfl_xor(d[3], 2, WAVE_SIZE);
-        const uint32_t r3 = __shfl_xor(d[2], 2, WAVE_SIZE);
+        const uint32_t r0 = dpp_xor2(d[1]);
+        const uint32_t r1 = dpp_xor2(d[0]);
+        const uint32_t r2 = dpp_xor2(d[3]);
+        const uint32_t r3 = dpp_xor2(d[2]);
         if (hi) {
             d[0] = r0;
             d[2] = r2;
@@ -81,13 +90,13 @@ __device__ __forceinline__ uint4 transpose8x8_bf16(uint4 v, int lane) {
             d[3] = r3;
         }
     }
-    // stage b=4: dword pairs (0<->2, 1<->3)
+    // stage (m=8, e=4): dword pairs (0<->2, 1<->3)
     {
-        const bool hi = (lane & 4) != 0;
-        const uint32_t r0 = __shfl_xor(d[2], 4, WAVE_SIZE);
-        const uint32_t r1 = __shfl_xor(d[3], 4, WAVE_SIZE);
-        const uint32_t r2 = __shfl_xor(d[0], 4, WAVE_SIZE);
-        const uint32_t r3 = __shfl_xor(d[1], 4, WAVE_SIZE);
+        const bool hi = (lane & 8) != 0;
+        const uint32_t r0 = dpp_xor8(d[2]);
+        const uint32_t r1 = dpp_xor8(d[3]);
+        const uint32_t r2 = dpp_xor8(d[0]);
+        const uint32_t r3 = dpp_xor8(d[1]);
         if (hi) {
             d[0] = r0;
             d[1] = r1;
@@ -147,7 +156,11 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
         const int cin0 = cint * CIN_T;
         // ---- stage weight fragments [tap][ks2][ct2][lane] into LDS -------
         {
-            for (int t = tid; t < NFRAG * WAVE_SIZE; t += NW * WAVE_SIZE) {
+            constexpr int WTRIP = (NFRAG + NW - 1) / NW;
+#pragma unroll
+            for (int it = 0; it < WTRIP; ++it) {
+                const int t = it * NW * WAVE_SIZE + tid;
+                if (NFRAG % NW != 0 && t >= NFRAG * WAVE_SIZE) break;
                 const int frag = t >> 6;
                 const int l = t & 63;
                 const int ct2 = frag % NCT;
@@ -174,7 +187,12 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
             constexpr int NXW = (XIN - 1 + 7) / 8;   // 8-x windows for xi>=1
             constexpr int NXS = (NXW + 7) / 8;       // windows per wave pass
             const bool w_vec = (p.W % 8) == 0;
-            for (int slab = wave; slab < YIN * CG * NXS; slab += NW) {
+            constexpr int NSLAB = YIN * CG * NXS;
+            constexpr int STRIP = (NSLAB + NW - 1) / NW;
+#pragma unroll
+            for (int sit = 0; sit < STRIP; ++sit) {
+                const int slab = sit * NW + wave;
+                if (NSLAB % NW != 0 && slab >= NSLAB) break;
                 const int xs = slab % NXS;
                 const int cg = (slab / NXS) % CG;
                 const int ry = slab / (NXS * CG);
@@ -193,8 +211,12 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                     src = bot;
                     sc = p.b_sc;
                 }
-                const int xw = xs * 8 + (lane >> 3);     // this lane's window
-                const int cin = cin0 + cg * 8 + (lane & 7);
+                // transpose-group member index (lane bits {0,1,3}) and
+                // window selector (bits {2,4,5}) — see transpose8x8_bf16
+                const int gr = (lane & 3) | ((lane & 8) >> 1);
+                const int gw = ((lane >> 2) & 1) | (((lane >> 4) & 3) << 1);
+                const int xw = xs * 8 + gw;              // this lane's window
+                const int cin = cin0 + cg * 8 + gr;
                 const int x_in0 = xb0 * S + xw * 8;      // xi = 1 + 8*xw
                 uint4 raw = {0, 0, 0, 0};
                 if (src && cin < p.Cin && 8 * xw + 1 < XIN) {
@@ -212,7 +234,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                     }
                 }
                 const uint4 tr = transpose8x8_bf16(raw, lane);
-                const int xi = 1 + xw * 8 + (lane & 7);  // lane now owns x=xi
+                const int xi = 1 + xw * 8 + gr;          // lane now owns x=xi
                 if (xi < XIN) {
                     const int plane = (S == 1) ? 0 : (xi & 1);
                     const int row = (S == 1) ? xi : (xi >> 1);
