@@ -231,6 +231,7 @@ at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
     TORCH_CHECK(p.KS * 16 >= Cin && p.CT * 32 >= cout, "packed weight too small");
     p.x_sb = x.stride(0);
     p.x_sc = x.stride(1);
+    if (const char* dbg = getenv("DFA_CONV_DEBUG")) p.debug = atoi(dbg);
     at::Tensor bias_c;
     if (bias.has_value()) {
         bias_c = bias->to(at::kBFloat16).contiguous();

@@ -177,6 +177,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 *reinterpret_cast<uint4*>(&w_lds[(int64_t)t * 16]) = w4;
             }
         }
+        if (p.debug != 1) {
         // ---- stage input tile [ry][xi][cin] transposed into LDS ----------
         // Each lane b128-loads 8 consecutive x of one cin row (coalesced
         // 128 B segments), an in-register 8x8 butterfly flips the slab to
@@ -265,8 +266,10 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 *reinterpret_cast<uint16_t*>(dst) = val;
             }
         }
+        }
         __syncthreads();
 
+        if (p.debug != 2) {
         // ---- accumulate 9 taps x KS_T k-slices ---------------------------
 #pragma unroll
         for (int tap = 0; tap < 9; ++tap) {
@@ -295,6 +298,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                             afrag, bfrag[px], acc[ct2][px], 0, 0, 0);
                 }
             }
+        }
         }
         __syncthreads();
     }

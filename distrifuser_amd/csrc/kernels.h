@@ -58,6 +58,7 @@ struct Conv3x3Params {
     uint16_t* o;
     int B, Cin, Cout, H, W, Ho, Wo;
     int KS, CT;
+    int debug;  // 1 = skip input staging, 2 = skip MFMA (phase-cost probes)
     int64_t x_sb, x_sc;
     int64_t t_sb, t_sc;
     int64_t b_sb, b_sc;
