@@ -111,11 +111,15 @@ __device__ __forceinline__ uint4 transpose8x8_bf16(uint4 v, int lane) {
 // S: conv stride; (YB rows) x (XW*32*PW x) output tile per block; CIN_T cins
 // per staged LDS tile; each wave owns PW 32-x subtiles x NCT 32-cout tiles
 // (A fragments reused across the PW subtiles, B across the NCT tiles).
-// Weight fragments for the (cin-tile, cout-block) are staged into LDS
-// cooperatively alongside the input tile: the v2 kernel loaded each A
-// fragment from L2 right before its MFMA and stalled on latency (PMC:
-// MFMA util 6%, SQ_WAIT dominated — see profiles/ notes); from LDS the
-// fragment read is a contiguous wave-wide 1 KB burst.
+//
+// Measured evolution (kernel_bench, 1280x1280@120^2, see profiles/):
+//   v1 scalar-gather staging ............ 120 TF (vmem instr bound)
+//   v3 LDS-staged weights + PW=2 ........ 385 TF (A-frag L2 latency fixed)
+//   v4 DPP transpose + unrolled staging .. 450 TF
+//   v5 (this): staging-volume cuts — NCT=4 halves the input re-read
+//   across cout-blocks; 8 B-padded LDS rows (stride 72 B, gcd(18,32)=2)
+//   make fragment reads 2-way instead of 8-way conflicted as b64 pairs;
+//   load bursts are issued in arrays before the transpose+write pass.
 template <int S, int YB, int XW, int CIN_T, int NCT, int PW>
 __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Params p) {
     constexpr int NW = YB * XW;
@@ -124,9 +128,9 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     constexpr int XIN = (XB - 1) * S + 3;          // staged input x positions
     constexpr int XP = (S == 1) ? XIN : (XB + 1);  // rows per x-parity plane
     constexpr int KS_T = CIN_T / 16;               // k-slices per staged tile
-    constexpr int ROW_B = CIN_T * 2;               // LDS row bytes
+    constexpr int ROW_P = CIN_T * 2 + 8;           // padded LDS row bytes
     constexpr int NFRAG = 9 * KS_T * NCT;          // weight frags per tile
-    __shared__ char in_lds[S * YIN * XP * ROW_B];
+    __shared__ char in_lds[S * YIN * XP * ROW_P];
     __shared__ char w_lds[NFRAG * WAVE_SIZE * 16];
 
     const int tid = threadIdx.x;
@@ -152,96 +156,126 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     // weights are packed to KS 16-cin slices (KS*16 a multiple of 64, zero
     // padded), so every staged tile is fully covered
     const int n_cin_tiles = (p.KS * 16 + CIN_T - 1) / CIN_T;
+    constexpr int WTRIP = (NFRAG + NW - 1) / NW;
+    constexpr int CG = CIN_T / 8;              // 8-cin groups
+    constexpr int NXW = (XIN - 1 + 7) / 8;     // 8-x windows for xi >= 1
+    constexpr int NXS = (NXW + 7) / 8;         // window strips per wave pass
+    constexpr int NSLAB = YIN * CG * NXS;
+    constexpr int STRIP = (NSLAB + NW - 1) / NW;
+
     for (int cint = 0; cint < n_cin_tiles; ++cint) {
         const int cin0 = cint * CIN_T;
-        // ---- stage weight fragments [tap][ks2][ct2][lane] into LDS -------
-        {
-            constexpr int WTRIP = (NFRAG + NW - 1) / NW;
+        // ---- staging: chunked load bursts -> transpose -> LDS writes -----
+        // (chunk sizes bound the live-register cost; full-tile arrays
+        // spilled 94 VGPRs at NCT=4)
+        const int gr = (lane & 3) | ((lane & 8) >> 1);
+        const int gw = ((lane >> 2) & 1) | (((lane >> 4) & 3) << 1);
+        const bool w_vec = (p.W % 8) == 0;
+        __syncthreads();  // previous tile's compute done: LDS reusable
+        if (p.debug != 3) {
+            constexpr int WCH = 3;  // weight-frag loads in flight per pass
 #pragma unroll
-            for (int it = 0; it < WTRIP; ++it) {
-                const int t = it * NW * WAVE_SIZE + tid;
-                if (NFRAG % NW != 0 && t >= NFRAG * WAVE_SIZE) break;
-                const int frag = t >> 6;
-                const int l = t & 63;
-                const int ct2 = frag % NCT;
-                const int ks2 = (frag / NCT) % KS_T;
-                const int tap = frag / (NCT * KS_T);
-                const int ct = cb * NCT + ct2;
-                uint4 w4 = {0, 0, 0, 0};
-                if (ct < p.CT) {
-                    const int64_t gidx =
-                        ((int64_t)(tap * p.KS + cint * KS_T + ks2) * p.CT + ct) * WAVE_SIZE + l;
-                    w4 = *reinterpret_cast<const uint4*>(p.wp + gidx * 8);
+            for (int c0 = 0; c0 < WTRIP; c0 += WCH) {
+                uint4 wraw[WCH];
+#pragma unroll
+                for (int j = 0; j < WCH; ++j) {
+                    const int it = c0 + j;
+                    wraw[j] = uint4{0, 0, 0, 0};
+                    if (it >= WTRIP) continue;
+                    const int t = it * NW * WAVE_SIZE + tid;
+                    if (NFRAG % NW == 0 || t < NFRAG * WAVE_SIZE) {
+                        const int frag = t >> 6;
+                        const int l = t & 63;
+                        const int ct2 = frag % NCT;
+                        const int ks2 = (frag / NCT) % KS_T;
+                        const int tap = frag / (NCT * KS_T);
+                        const int ct = cb * NCT + ct2;
+                        if (ct < p.CT) {
+                            const int64_t gidx =
+                                ((int64_t)(tap * p.KS + cint * KS_T + ks2) * p.CT + ct) *
+                                    WAVE_SIZE + l;
+                            wraw[j] = *reinterpret_cast<const uint4*>(p.wp + gidx * 8);
+                        }
+                    }
                 }
-                // LDS layout mirrors the compute loop: [tap][ks2][ct2][l]
-                *reinterpret_cast<uint4*>(&w_lds[(int64_t)t * 16]) = w4;
+#pragma unroll
+                for (int j = 0; j < WCH; ++j) {
+                    const int it = c0 + j;
+                    if (it >= WTRIP) continue;
+                    const int t = it * NW * WAVE_SIZE + tid;
+                    if (NFRAG % NW == 0 || t < NFRAG * WAVE_SIZE)
+                        *reinterpret_cast<uint4*>(&w_lds[(int64_t)t * 16]) = wraw[j];
+                }
             }
         }
         if (p.debug != 1) {
-        // ---- stage input tile [ry][xi][cin] transposed into LDS ----------
-        // Each lane b128-loads 8 consecutive x of one cin row (coalesced
-        // 128 B segments), an in-register 8x8 butterfly flips the slab to
-        // [x][cin], one b128 LDS write per lane. Edges / W%8!=0 take
-        // guarded scalar element loads inside the same slab.
-        {
-            constexpr int CG = CIN_T / 8;            // 8-cin groups
-            constexpr int NXW = (XIN - 1 + 7) / 8;   // 8-x windows for xi>=1
-            constexpr int NXS = (NXW + 7) / 8;       // windows per wave pass
-            const bool w_vec = (p.W % 8) == 0;
-            constexpr int NSLAB = YIN * CG * NXS;
-            constexpr int STRIP = (NSLAB + NW - 1) / NW;
+            constexpr int ICH = 5;  // input slab loads in flight per pass
 #pragma unroll
-            for (int sit = 0; sit < STRIP; ++sit) {
-                const int slab = sit * NW + wave;
-                if (NSLAB % NW != 0 && slab >= NSLAB) break;
-                const int xs = slab % NXS;
-                const int cg = (slab / NXS) % CG;
-                const int ry = slab / (NXS * CG);
-                const int y_in = yb0 * S - 1 + ry;
-                const uint16_t* src = nullptr;
-                int64_t sc = 0;
-                int64_t rbase = 0;
-                if (y_in >= 0 && y_in < p.H) {
-                    src = xin;
-                    sc = p.x_sc;
-                    rbase = (int64_t)y_in * p.W;
-                } else if (y_in == -1 && top) {
-                    src = top;
-                    sc = p.t_sc;
-                } else if (y_in == p.H && bot) {
-                    src = bot;
-                    sc = p.b_sc;
-                }
-                // transpose-group member index (lane bits {0,1,3}) and
-                // window selector (bits {2,4,5}) — see transpose8x8_bf16
-                const int gr = (lane & 3) | ((lane & 8) >> 1);
-                const int gw = ((lane >> 2) & 1) | (((lane >> 4) & 3) << 1);
-                const int xw = xs * 8 + gw;              // this lane's window
-                const int cin = cin0 + cg * 8 + gr;
-                const int x_in0 = xb0 * S + xw * 8;      // xi = 1 + 8*xw
-                uint4 raw = {0, 0, 0, 0};
-                if (src && cin < p.Cin && 8 * xw + 1 < XIN) {
-                    const uint16_t* rp = src + (int64_t)cin * sc + rbase;
-                    if (w_vec && x_in0 + 8 <= p.W) {
-                        raw = *reinterpret_cast<const uint4*>(rp + x_in0);
-                    } else {
-                        uint16_t vals[8];
+            for (int c0 = 0; c0 < STRIP; c0 += ICH) {
+                uint4 iraw[ICH];
 #pragma unroll
-                        for (int j = 0; j < 8; ++j) {
-                            const int x_in = x_in0 + j;
-                            vals[j] = (x_in < p.W) ? rp[x_in] : (uint16_t)0;
+                for (int j = 0; j < ICH; ++j) {
+                    const int sit = c0 + j;
+                    iraw[j] = uint4{0, 0, 0, 0};
+                    if (sit >= STRIP) continue;
+                    const int slab = sit * NW + wave;
+                    if (NSLAB % NW != 0 && slab >= NSLAB) continue;
+                    const int xs = slab % NXS;
+                    const int cg = (slab / NXS) % CG;
+                    const int ry = slab / (NXS * CG);
+                    const int y_in = yb0 * S - 1 + ry;
+                    const uint16_t* src = nullptr;
+                    int64_t sc = 0;
+                    int64_t rbase = 0;
+                    if (y_in >= 0 && y_in < p.H) {
+                        src = xin;
+                        sc = p.x_sc;
+                        rbase = (int64_t)y_in * p.W;
+                    } else if (y_in == -1 && top) {
+                        src = top;
+                        sc = p.t_sc;
+                    } else if (y_in == p.H && bot) {
+                        src = bot;
+                        sc = p.b_sc;
+                    }
+                    const int xw = xs * 8 + gw;
+                    const int cin = cin0 + cg * 8 + gr;
+                    const int x_in0 = xb0 * S + xw * 8;  // covers xi = 1+8*xw
+                    if (src && cin < p.Cin && 8 * xw + 1 < XIN) {
+                        const uint16_t* rp = src + (int64_t)cin * sc + rbase;
+                        if (w_vec && x_in0 + 8 <= p.W) {
+                            iraw[j] = *reinterpret_cast<const uint4*>(rp + x_in0);
+                        } else {
+                            uint16_t vals[8];
+#pragma unroll
+                            for (int jj = 0; jj < 8; ++jj) {
+                                const int x_in = x_in0 + jj;
+                                vals[jj] = (x_in < p.W) ? rp[x_in] : (uint16_t)0;
+                            }
+                            iraw[j] = *reinterpret_cast<const uint4*>(vals);
                         }
-                        raw = *reinterpret_cast<const uint4*>(vals);
                     }
                 }
-                const uint4 tr = transpose8x8_bf16(raw, lane);
-                const int xi = 1 + xw * 8 + gr;          // lane now owns x=xi
-                if (xi < XIN) {
-                    const int plane = (S == 1) ? 0 : (xi & 1);
-                    const int row = (S == 1) ? xi : (xi >> 1);
-                    char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
-                                        cv_swz<ROW_B>(row, cg * 16)];
-                    *reinterpret_cast<uint4*>(dst) = tr;
+#pragma unroll
+                for (int j = 0; j < ICH; ++j) {
+                    const int sit = c0 + j;
+                    if (sit >= STRIP) continue;
+                    const int slab = sit * NW + wave;
+                    if (NSLAB % NW != 0 && slab >= NSLAB) continue;
+                    const int xs = slab % NXS;
+                    const int cg = (slab / NXS) % CG;
+                    const int ry = slab / (NXS * CG);
+                    const int xw = xs * 8 + gw;
+                    const uint4 tr = transpose8x8_bf16(iraw[j], lane);
+                    const int xi = 1 + xw * 8 + gr;  // lane now owns x = xi
+                    if (xi < XIN) {
+                        const int plane = (S == 1) ? 0 : (xi & 1);
+                        const int row = (S == 1) ? xi : (xi >> 1);
+                        char* dst =
+                            &in_lds[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
+                        *reinterpret_cast<uint2*>(dst) = uint2{tr.x, tr.y};
+                        *reinterpret_cast<uint2*>(dst + 8) = uint2{tr.z, tr.w};
+                    }
                 }
             }
             // left halo column xi = 0 (x_in = xb0*S - 1): scalar, tiny
@@ -260,45 +294,46 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                     else if (y_in == p.H && bot)
                         val = bot[(int64_t)cin * p.b_sc + x_in];
                 }
-                const int row = 0;  // xi = 0: plane 0, row 0 for both strides
-                char* dst = &in_lds[((0 * YIN + ry) * XP + row) * ROW_B +
-                                    cv_swz<ROW_B>(row, (ci / 8) * 16) + (ci % 8) * 2];
-                *reinterpret_cast<uint16_t*>(dst) = val;
+                // xi = 0: plane 0, row 0 for both strides
+                *reinterpret_cast<uint16_t*>(
+                    &in_lds[((0 * YIN + ry) * XP + 0) * ROW_P + ci * 2]) = val;
             }
-        }
         }
         __syncthreads();
 
         if (p.debug != 2) {
-        // ---- accumulate 9 taps x KS_T k-slices ---------------------------
+            // ---- accumulate 9 taps x KS_T k-slices -----------------------
 #pragma unroll
-        for (int tap = 0; tap < 9; ++tap) {
-            const int dy = tap / 3;
-            const int dx = tap % 3;
-            const int ry = wy * S + dy;
+            for (int tap = 0; tap < 9; ++tap) {
+                const int dy = tap / 3;
+                const int dx = tap % 3;
+                const int ry = wy * S + dy;
 #pragma unroll
-            for (int ks2 = 0; ks2 < KS_T; ++ks2) {
-                short8 bfrag[PW];
+                for (int ks2 = 0; ks2 < KS_T; ++ks2) {
+                    short8 bfrag[PW];
 #pragma unroll
-                for (int px = 0; px < PW; ++px) {
-                    const int xi = ((wx * PW + px) * 32 + lo) * S + dx;
-                    const int plane = (S == 1) ? 0 : (xi & 1);
-                    const int row = (S == 1) ? xi : (xi >> 1);
-                    bfrag[px] = *reinterpret_cast<const short8*>(
-                        &in_lds[((plane * YIN + ry) * XP + row) * ROW_B +
-                                cv_swz<ROW_B>(row, (ks2 * 16 + hi * 8) * 2)]);
-                }
+                    for (int px = 0; px < PW; ++px) {
+                        const int xi = ((wx * PW + px) * 32 + lo) * S + dx;
+                        const int plane = (S == 1) ? 0 : (xi & 1);
+                        const int row = (S == 1) ? xi : (xi >> 1);
+                        const char* baddr = &in_lds[((plane * YIN + ry) * XP + row) * ROW_P +
+                                                    (ks2 * 16 + hi * 8) * 2];
+                        const uint2 b0 = *reinterpret_cast<const uint2*>(baddr);
+                        const uint2 b1 = *reinterpret_cast<const uint2*>(baddr + 8);
+                        const uint4 bb{b0.x, b0.y, b1.x, b1.y};
+                        bfrag[px] = __builtin_bit_cast(short8, bb);
+                    }
 #pragma unroll
-                for (int ct2 = 0; ct2 < NCT; ++ct2) {
-                    const short8 afrag = *reinterpret_cast<const short8*>(
-                        &w_lds[(((tap * KS_T + ks2) * NCT + ct2) * WAVE_SIZE + lane) * 16]);
+                    for (int ct2 = 0; ct2 < NCT; ++ct2) {
+                        const short8 afrag = *reinterpret_cast<const short8*>(
+                            &w_lds[(((tap * KS_T + ks2) * NCT + ct2) * WAVE_SIZE + lane) * 16]);
 #pragma unroll
-                    for (int px = 0; px < PW; ++px)
-                        acc[ct2][px] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                            afrag, bfrag[px], acc[ct2][px], 0, 0, 0);
+                        for (int px = 0; px < PW; ++px)
+                            acc[ct2][px] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                                afrag, bfrag[px], acc[ct2][px], 0, 0, 0);
+                    }
                 }
             }
-        }
         }
         __syncthreads();
     }
@@ -333,17 +368,18 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
 
 void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
     if (stride == 1) {
-        // 8 waves: 8 output rows x 1 x-wave of 64 px (PW=2 A-frag reuse);
-        // LDS: input 10x66x128 B = 84.5 KB + 72 KB weight frags = 156.5 KB.
-        constexpr int YB = 8, XW = 1, NCT = 2, PW = 2;
+        // 8 waves: 8 output rows x 1 x-wave of 64 px (PW=2 A-frag reuse),
+        // 128 couts per block (NCT=4 halves the input re-read);
+        // LDS: input 10x66x72 B = 46.4 KB + 72 KB weight frags.
+        constexpr int YB = 8, XW = 1, NCT = 4, PW = 2;
         constexpr int XB = XW * 32 * PW;
         const int nxb = (p.Wo + XB - 1) / XB;
         const int nyb = (p.Ho + YB - 1) / YB;
         dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
-        conv3x3_kernel<1, YB, XW, 64, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+        conv3x3_kernel<1, YB, XW, 32, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     } else {
         // stride 2: 8 waves: 4 output rows x 2 x-waves of 32 px;
-        // LDS: parity-plane input 74.9 KB + 36 KB weight frags.
+        // LDS: parity-plane input 2x9x65x72 B = 84.2 KB + 36 KB weights.
         constexpr int YB = 4, XW = 2, NCT = 2, PW = 1;
         constexpr int XB = XW * 32 * PW;
         const int nxb = (p.Wo + XB - 1) / XB;
