@@ -141,11 +141,24 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     const int wy = wave / XW;
     const int wx = wave % XW;
 
+    // XCD-aware decode of the 1-D grid: hardware assigns dispatch D to XCD
+    // D%8, so giving the NCB cout-blocks of one pixel-tile consecutive D/8
+    // slots at the same D%8 keeps them CONCURRENT ON ONE XCD — the staged
+    // input tile and the cout-block's weights then live in that XCD's L2.
+    // (Without this, PMC showed every staged byte fetched from HBM:
+    // TCC_EA_RDREQ == staged volume, 1.45 GB/dispatch at 1280^2@120^2.)
     const int nxb = (p.Wo + XB - 1) / XB;
-    const int xb0 = (blockIdx.x % nxb) * XB;
-    const int yb0 = (blockIdx.x / nxb) * YB;
-    const int cb = blockIdx.y;
-    const int b = blockIdx.z;
+    const int nyb = (p.Ho + YB - 1) / YB;
+    const int npix = nxb * nyb * p.B;  // pixel-tiles x batch
+    const int ncb = (p.CT + NCT - 1) / NCT;
+    const int D = blockIdx.x;
+    const int cb = (D >> 3) % ncb;
+    const int pp = (D & 7) + 8 * ((D >> 3) / ncb);
+    if (pp >= npix) return;
+    const int b = pp / (nxb * nyb);
+    const int pt = pp % (nxb * nyb);
+    const int xb0 = (pt % nxb) * XB;
+    const int yb0 = (pt / nxb) * YB;
 
     const uint16_t* xin = p.x + (int64_t)b * p.x_sb;
     const uint16_t* top = p.top ? p.top + (int64_t)b * p.t_sb : nullptr;
@@ -375,7 +388,9 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
         constexpr int XB = XW * 32 * PW;
         const int nxb = (p.Wo + XB - 1) / XB;
         const int nyb = (p.Ho + YB - 1) / YB;
-        dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
+        const int npix = nxb * nyb * p.B;
+        const int ncb = (p.CT + NCT - 1) / NCT;
+        dim3 grid((unsigned)(8 * ncb * ((npix + 7) / 8)));
         conv3x3_kernel<1, YB, XW, 32, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     } else {
         // stride 2: 8 waves: 4 output rows x 2 x-waves of 32 px;
@@ -384,7 +399,9 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
         constexpr int XB = XW * 32 * PW;
         const int nxb = (p.Wo + XB - 1) / XB;
         const int nyb = (p.Ho + YB - 1) / YB;
-        dim3 grid((unsigned)(nxb * nyb), (unsigned)((p.CT + NCT - 1) / NCT), (unsigned)p.B);
+        const int npix = nxb * nyb * p.B;
+        const int ncb = (p.CT + NCT - 1) / NCT;
+        dim3 grid((unsigned)(8 * ncb * ((npix + 7) / 8)));
         conv3x3_kernel<2, YB, XW, 32, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     }
 }
