@@ -36,6 +36,23 @@
 namespace {
 
 typedef float float16v __attribute__((ext_vector_type(16)));
+typedef int int32x4_ck __attribute__((ext_vector_type(4)));
+
+// raw-buffer load with hardware bounds check (OOB -> 0); gfx9 descriptor
+// word3 = 0x00020000 (same constant CK uses, /opt/rocm/include/ck/ck.hpp:82)
+__device__ int32x4_ck llvm_amdgcn_raw_buffer_load_x4(
+    int32x4_ck srsrc, int voffset, int soffset,
+    int glc_slc) __asm("llvm.amdgcn.raw.buffer.load.v4i32");
+
+__device__ __forceinline__ int32x4_ck make_srsrc(const void* base, uint32_t bytes) {
+    const uint64_t a = (uint64_t)base;
+    int32x4_ck r;
+    r.x = (int)(uint32_t)a;
+    r.y = (int)(uint32_t)(a >> 32);
+    r.z = (int)bytes;
+    r.w = 0x00020000;
+    return r;
+}
 
 template <int ROW_BYTES>
 __device__ __forceinline__ int cv_swz(int row, int byte_off) {
@@ -176,14 +193,40 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     constexpr int NSLAB = YIN * CG * NXS;
     constexpr int STRIP = (NSLAB + NW - 1) / NW;
 
+    // ---- v7 staging prologue: per-slab state hoisted out of the k-loop.
+    // Interior slabs load through a buffer descriptor whose num_records
+    // (= Cin * sc * 2 B) zeroes the cin tail in hardware, so the hot loop
+    // has no per-lane masks and no 64-bit address math (the v6 staging asm
+    // was saveexec/mad_i64-bound: ~21 us per 120 KB tile).
+    const int gr = (lane & 3) | ((lane & 8) >> 1);
+    const int gw = ((lane >> 2) & 1) | (((lane >> 4) & 3) << 1);
+    const int32x4_ck desc0 = make_srsrc(xin, (uint32_t)((int64_t)p.Cin * p.x_sc * 2));
+    const uint32_t voff_inc = (uint32_t)(CIN_T * p.x_sc * 2);
+    const bool w_vec = (p.W % 8) == 0;
+    uint32_t voff[STRIP];
+    uint32_t interior_mask = 0, tail_mask = 0, valid_mask = 0;
+#pragma unroll
+    for (int sit = 0; sit < STRIP; ++sit) {
+        const int slab = sit * NW + wave;
+        voff[sit] = 0;
+        if (NSLAB % NW != 0 && slab >= NSLAB) continue;
+        valid_mask |= 1u << sit;
+        const int xs = slab % NXS;
+        const int cg = (slab / NXS) % CG;
+        const int ry = slab / (NXS * CG);
+        const int y_in = yb0 * S - 1 + ry;
+        const int xw = xs * 8 + gw;
+        const int x_in0 = xb0 * S + xw * 8;
+        if (y_in >= 0 && y_in < p.H) interior_mask |= 1u << sit;
+        // does any lane's window cross the right image edge? (x tail)
+        if (xb0 * S + (xs * 8 + 7) * 8 + 8 > p.W) tail_mask |= 1u << sit;
+        voff[sit] = (uint32_t)(((int64_t)(cg * 8 + gr) * p.x_sc +
+                                (int64_t)(y_in >= 0 ? y_in : 0) * p.W + x_in0) * 2);
+    }
+
     for (int cint = 0; cint < n_cin_tiles; ++cint) {
         const int cin0 = cint * CIN_T;
-        // ---- staging: chunked load bursts -> transpose -> LDS writes -----
-        // (chunk sizes bound the live-register cost; full-tile arrays
-        // spilled 94 VGPRs at NCT=4)
-        const int gr = (lane & 3) | ((lane & 8) >> 1);
-        const int gw = ((lane >> 2) & 1) | (((lane >> 4) & 3) << 1);
-        const bool w_vec = (p.W % 8) == 0;
+        // ---- staging: weights chunked; input via buffer-descriptor slabs --
         __syncthreads();  // previous tile's compute done: LDS reusable
         if (p.debug != 3) {
             constexpr int WCH = 3;  // weight-frag loads in flight per pass
@@ -222,75 +265,80 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
             }
         }
         if (p.debug != 1) {
-            constexpr int ICH = 5;  // input slab loads in flight per pass
 #pragma unroll
-            for (int c0 = 0; c0 < STRIP; c0 += ICH) {
-                uint4 iraw[ICH];
+            for (int sit = 0; sit < STRIP; ++sit) {
+                if (!((valid_mask >> sit) & 1)) continue;
+                const int slab = sit * NW + wave;
+                const int xs = slab % NXS;
+                const int cg = (slab / NXS) % CG;
+                const int ry = slab / (NXS * CG);
+                const int xw = xs * 8 + gw;
+                const int x_in0 = xb0 * S + xw * 8;
+                uint4 raw = {0, 0, 0, 0};
+                if ((interior_mask >> sit) & 1) {
+                    if (w_vec) {
+                        raw = __builtin_bit_cast(
+                            uint4, llvm_amdgcn_raw_buffer_load_x4(desc0, (int)voff[sit], 0, 0));
+                        if ((tail_mask >> sit) & 1) {
+                            // zero the x >= W elements of this lane's window
+                            const int rem = p.W - x_in0;  // may be <= 0
+                            uint32_t* dw = reinterpret_cast<uint32_t*>(&raw);
 #pragma unroll
-                for (int j = 0; j < ICH; ++j) {
-                    const int sit = c0 + j;
-                    iraw[j] = uint4{0, 0, 0, 0};
-                    if (sit >= STRIP) continue;
-                    const int slab = sit * NW + wave;
-                    if (NSLAB % NW != 0 && slab >= NSLAB) continue;
-                    const int xs = slab % NXS;
-                    const int cg = (slab / NXS) % CG;
-                    const int ry = slab / (NXS * CG);
+                            for (int d = 0; d < 4; ++d) {
+                                const uint32_t m = (rem >= 2 * d + 2)
+                                                       ? 0xffffffffu
+                                                       : ((rem == 2 * d + 1) ? 0x0000ffffu : 0u);
+                                dw[d] &= m;
+                            }
+                        }
+                    } else {
+                        // W % 8 != 0: rows are not 16 B aligned; guarded
+                        // scalar loads (correctness path, non-SD shapes)
+                        const int cin = cin0 + cg * 8 + gr;
+                        const int y_in = yb0 * S - 1 + ry;
+                        if (cin < p.Cin) {
+                            const uint16_t* rp = xin + (int64_t)cin * p.x_sc + (int64_t)y_in * p.W;
+                            uint16_t vals[8];
+#pragma unroll
+                            for (int jj = 0; jj < 8; ++jj)
+                                vals[jj] = (x_in0 + jj < p.W) ? rp[x_in0 + jj] : (uint16_t)0;
+                            raw = *reinterpret_cast<const uint4*>(vals);
+                        }
+                    }
+                } else {
+                    // halo / zero rows (block at a patch boundary): rare
                     const int y_in = yb0 * S - 1 + ry;
                     const uint16_t* src = nullptr;
                     int64_t sc = 0;
-                    int64_t rbase = 0;
-                    if (y_in >= 0 && y_in < p.H) {
-                        src = xin;
-                        sc = p.x_sc;
-                        rbase = (int64_t)y_in * p.W;
-                    } else if (y_in == -1 && top) {
+                    if (y_in == -1 && top) {
                         src = top;
                         sc = p.t_sc;
                     } else if (y_in == p.H && bot) {
                         src = bot;
                         sc = p.b_sc;
                     }
-                    const int xw = xs * 8 + gw;
                     const int cin = cin0 + cg * 8 + gr;
-                    const int x_in0 = xb0 * S + xw * 8;  // covers xi = 1+8*xw
-                    if (src && cin < p.Cin && 8 * xw + 1 < XIN) {
-                        const uint16_t* rp = src + (int64_t)cin * sc + rbase;
-                        if (w_vec && x_in0 + 8 <= p.W) {
-                            iraw[j] = *reinterpret_cast<const uint4*>(rp + x_in0);
-                        } else {
-                            uint16_t vals[8];
+                    if (src && cin < p.Cin) {
+                        const uint16_t* rp = src + (int64_t)cin * sc;
+                        uint16_t vals[8];
 #pragma unroll
-                            for (int jj = 0; jj < 8; ++jj) {
-                                const int x_in = x_in0 + jj;
-                                vals[jj] = (x_in < p.W) ? rp[x_in] : (uint16_t)0;
-                            }
-                            iraw[j] = *reinterpret_cast<const uint4*>(vals);
-                        }
+                        for (int jj = 0; jj < 8; ++jj)
+                            vals[jj] = (x_in0 + jj < p.W) ? rp[x_in0 + jj] : (uint16_t)0;
+                        raw = *reinterpret_cast<const uint4*>(vals);
                     }
                 }
-#pragma unroll
-                for (int j = 0; j < ICH; ++j) {
-                    const int sit = c0 + j;
-                    if (sit >= STRIP) continue;
-                    const int slab = sit * NW + wave;
-                    if (NSLAB % NW != 0 && slab >= NSLAB) continue;
-                    const int xs = slab % NXS;
-                    const int cg = (slab / NXS) % CG;
-                    const int ry = slab / (NXS * CG);
-                    const int xw = xs * 8 + gw;
-                    const uint4 tr = transpose8x8_bf16(iraw[j], lane);
-                    const int xi = 1 + xw * 8 + gr;  // lane now owns x = xi
-                    if (xi < XIN) {
-                        const int plane = (S == 1) ? 0 : (xi & 1);
-                        const int row = (S == 1) ? xi : (xi >> 1);
-                        char* dst =
-                            &in_lds[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
-                        *reinterpret_cast<uint2*>(dst) = uint2{tr.x, tr.y};
-                        *reinterpret_cast<uint2*>(dst + 8) = uint2{tr.z, tr.w};
-                    }
+                const uint4 tr = transpose8x8_bf16(raw, lane);
+                const int xi = 1 + xw * 8 + gr;  // lane now owns x = xi
+                if (xi < XIN) {
+                    const int plane = (S == 1) ? 0 : (xi & 1);
+                    const int row = (S == 1) ? xi : (xi >> 1);
+                    char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
+                    *reinterpret_cast<uint2*>(dst) = uint2{tr.x, tr.y};
+                    *reinterpret_cast<uint2*>(dst + 8) = uint2{tr.z, tr.w};
                 }
             }
+#pragma unroll
+            for (int sit = 0; sit < STRIP; ++sit) voff[sit] += voff_inc;
             // left halo column xi = 0 (x_in = xb0*S - 1): scalar, tiny
             for (int c = tid; c < YIN * CIN_T; c += NW * WAVE_SIZE) {
                 const int ci = c % CIN_T;
