@@ -147,8 +147,13 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     constexpr int KS_T = CIN_T / 16;               // k-slices per staged tile
     constexpr int ROW_P = CIN_T * 2 + 8;           // padded LDS row bytes
     constexpr int NFRAG = 9 * KS_T * NCT;          // weight frags per tile
-    __shared__ char in_lds[S * YIN * XP * ROW_P];
-    __shared__ char w_lds[NFRAG * WAVE_SIZE * 16];
+    constexpr int IN_SZ = S * YIN * XP * ROW_P;
+    constexpr int W_SZ = NFRAG * WAVE_SIZE * 16;
+    // double-buffered: tile t+1 stages into buffer (t+1)&1 while tile t
+    // computes from buffer t&1 (one barrier per tile; the staging loads'
+    // HBM/L2 latency hides under the MFMA phase)
+    __shared__ char in_lds[2 * IN_SZ];
+    __shared__ char w_lds[2 * W_SZ];
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
@@ -224,10 +229,8 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                                 (int64_t)(y_in >= 0 ? y_in : 0) * p.W + x_in0) * 2);
     }
 
-    for (int cint = 0; cint < n_cin_tiles; ++cint) {
+    auto stage_tile = [&](int cint, char* inb, char* wb) {
         const int cin0 = cint * CIN_T;
-        // ---- staging: weights chunked; input via buffer-descriptor slabs --
-        __syncthreads();  // previous tile's compute done: LDS reusable
         if (p.debug != 3) {
             constexpr int WCH = 3;  // weight-frag loads in flight per pass
 #pragma unroll
@@ -260,7 +263,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                     if (it >= WTRIP) continue;
                     const int t = it * NW * WAVE_SIZE + tid;
                     if (NFRAG % NW == 0 || t < NFRAG * WAVE_SIZE)
-                        *reinterpret_cast<uint4*>(&w_lds[(int64_t)t * 16]) = wraw[j];
+                        *reinterpret_cast<uint4*>(&wb[(int64_t)t * 16]) = wraw[j];
                 }
             }
         }
@@ -332,7 +335,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 if (xi < XIN) {
                     const int plane = (S == 1) ? 0 : (xi & 1);
                     const int row = (S == 1) ? xi : (xi >> 1);
-                    char* dst = &in_lds[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
+                    char* dst = &inb[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
                     *reinterpret_cast<uint2*>(dst) = uint2{tr.x, tr.y};
                     *reinterpret_cast<uint2*>(dst + 8) = uint2{tr.z, tr.w};
                 }
@@ -357,11 +360,11 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 }
                 // xi = 0: plane 0, row 0 for both strides
                 *reinterpret_cast<uint16_t*>(
-                    &in_lds[((0 * YIN + ry) * XP + 0) * ROW_P + ci * 2]) = val;
+                    &inb[((0 * YIN + ry) * XP + 0) * ROW_P + ci * 2]) = val;
             }
         }
-        __syncthreads();
-
+    };
+    auto compute_tile = [&](char* inb, char* wb) {
         if (p.debug != 2) {
             // ---- accumulate 9 taps x KS_T k-slices -----------------------
 #pragma unroll
@@ -377,7 +380,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                         const int xi = ((wx * PW + px) * 32 + lo) * S + dx;
                         const int plane = (S == 1) ? 0 : (xi & 1);
                         const int row = (S == 1) ? xi : (xi >> 1);
-                        const char* baddr = &in_lds[((plane * YIN + ry) * XP + row) * ROW_P +
+                        const char* baddr = &inb[((plane * YIN + ry) * XP + row) * ROW_P +
                                                     (ks2 * 16 + hi * 8) * 2];
                         const uint2 b0 = *reinterpret_cast<const uint2*>(baddr);
                         const uint2 b1 = *reinterpret_cast<const uint2*>(baddr + 8);
@@ -387,7 +390,7 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
 #pragma unroll
                     for (int ct2 = 0; ct2 < NCT; ++ct2) {
                         const short8 afrag = *reinterpret_cast<const short8*>(
-                            &w_lds[(((tap * KS_T + ks2) * NCT + ct2) * WAVE_SIZE + lane) * 16]);
+                            &wb[(((tap * KS_T + ks2) * NCT + ct2) * WAVE_SIZE + lane) * 16]);
 #pragma unroll
                         for (int px = 0; px < PW; ++px)
                             acc[ct2][px] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -396,7 +399,17 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 }
             }
         }
-        __syncthreads();
+    };
+
+    if (n_cin_tiles > 0) stage_tile(0, in_lds, w_lds);
+    for (int cint = 0; cint < n_cin_tiles; ++cint) {
+        char* inb = in_lds + (cint & 1) * IN_SZ;
+        char* wb = w_lds + (cint & 1) * W_SZ;
+        __syncthreads();  // staged tile cint visible; buffer cint+1 free
+        if (cint + 1 < n_cin_tiles)
+            stage_tile(cint + 1, in_lds + ((cint + 1) & 1) * IN_SZ,
+                       w_lds + ((cint + 1) & 1) * W_SZ);
+        compute_tile(inb, wb);
     }
 
     // ---- epilogue: O[b][cout][y][x] = acc + bias ---------------------------
@@ -431,7 +444,7 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
     if (stride == 1) {
         // 8 waves: 8 output rows x 1 x-wave of 64 px (PW=2 A-frag reuse),
         // 128 couts per block (NCT=4 halves the input re-read);
-        // LDS: input 10x66x72 B = 46.4 KB + 72 KB weight frags.
+        // LDS (double-buffered): 2 x (input 10x66x40 B + 36 KB weight frags).
         constexpr int YB = 8, XW = 1, NCT = 4, PW = 2;
         constexpr int XB = XW * 32 * PW;
         const int nxb = (p.Wo + XB - 1) / XB;
@@ -439,10 +452,10 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
         const int npix = nxb * nyb * p.B;
         const int ncb = (p.CT + NCT - 1) / NCT;
         dim3 grid((unsigned)(8 * ncb * ((npix + 7) / 8)));
-        conv3x3_kernel<1, YB, XW, 32, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+        conv3x3_kernel<1, YB, XW, 16, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     } else {
         // stride 2: 8 waves: 4 output rows x 2 x-waves of 32 px;
-        // LDS: parity-plane input 2x9x65x72 B = 84.2 KB + 36 KB weights.
+        // LDS (double-buffered): 2 x (parity input 46.8 KB + 18.4 KB weights).
         constexpr int YB = 4, XW = 2, NCT = 2, PW = 1;
         constexpr int XB = XW * 32 * PW;
         const int nxb = (p.Wo + XB - 1) / XB;
@@ -450,6 +463,6 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
         const int npix = nxb * nyb * p.B;
         const int ncb = (p.CT + NCT - 1) / NCT;
         dim3 grid((unsigned)(8 * ncb * ((npix + 7) / 8)));
-        conv3x3_kernel<2, YB, XW, 32, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
+        conv3x3_kernel<2, YB, XW, 16, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     }
 }
