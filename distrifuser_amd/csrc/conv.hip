@@ -229,100 +229,86 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                                 (int64_t)(y_in >= 0 ? y_in : 0) * p.W + x_in0) * 2);
     }
 
-    auto stage_tile = [&](int cint, char* inb, char* wb) {
-        const int cin0 = cint * CIN_T;
-        if (p.debug != 3) {
-            constexpr int WCH = 3;  // weight-frag loads in flight per pass
+    // ---- v8 pipeline phases ------------------------------------------------
+    // Per tile t (while tile t-1 computes): weights stream straight to LDS
+    // via global_load_lds DMA (no registers, wave-uniform LDS base +
+    // lane*16 matches the linear fragment layout exactly); input slabs are
+    // buffer-loaded into registers and written to LDS AFTER the compute
+    // phase (T14 issue-early/write-late — __syncthreads drains vmcnt(0), so
+    // in-flight loads cannot cross a barrier and the only overlap window is
+    // within the iteration).
+    auto stage_weights = [&](int cint, char* wb) {
+        if (p.debug == 3) return;
 #pragma unroll
-            for (int c0 = 0; c0 < WTRIP; c0 += WCH) {
-                uint4 wraw[WCH];
-#pragma unroll
-                for (int j = 0; j < WCH; ++j) {
-                    const int it = c0 + j;
-                    wraw[j] = uint4{0, 0, 0, 0};
-                    if (it >= WTRIP) continue;
-                    const int t = it * NW * WAVE_SIZE + tid;
-                    if (NFRAG % NW == 0 || t < NFRAG * WAVE_SIZE) {
-                        const int frag = t >> 6;
-                        const int l = t & 63;
-                        const int ct2 = frag % NCT;
-                        const int ks2 = (frag / NCT) % KS_T;
-                        const int tap = frag / (NCT * KS_T);
-                        const int ct = cb * NCT + ct2;
-                        if (ct < p.CT) {
-                            const int64_t gidx =
-                                ((int64_t)(tap * p.KS + cint * KS_T + ks2) * p.CT + ct) *
-                                    WAVE_SIZE + l;
-                            wraw[j] = *reinterpret_cast<const uint4*>(p.wp + gidx * 8);
-                        }
-                    }
-                }
-#pragma unroll
-                for (int j = 0; j < WCH; ++j) {
-                    const int it = c0 + j;
-                    if (it >= WTRIP) continue;
-                    const int t = it * NW * WAVE_SIZE + tid;
-                    if (NFRAG % NW == 0 || t < NFRAG * WAVE_SIZE)
-                        *reinterpret_cast<uint4*>(&wb[(int64_t)t * 16]) = wraw[j];
-                }
+        for (int it = 0; it < WTRIP; ++it) {
+            const int frag = it * NW + wave;
+            if (NFRAG % NW != 0 && frag >= NFRAG) break;
+            const int ct2 = frag % NCT;
+            const int ks2 = (frag / NCT) % KS_T;
+            const int tap = frag / (NCT * KS_T);
+            const int ct = cb * NCT + ct2;
+            char* ldst = wb + (int64_t)frag * WAVE_SIZE * 16;
+            if (ct < p.CT) {
+                const uint16_t* g =
+                    p.wp +
+                    (((int64_t)(tap * p.KS + cint * KS_T + ks2) * p.CT + ct) * WAVE_SIZE + lane) *
+                        8;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) uint32_t*)g,
+                    (__attribute__((address_space(3))) uint32_t*)ldst, 16, 0, 0);
+            } else {
+                *reinterpret_cast<uint4*>(ldst + lane * 16) = uint4{0, 0, 0, 0};
             }
         }
-        if (p.debug != 1) {
+    };
+    uint4 iregs[STRIP];
+    auto load_input = [&]() {
+        if (p.debug == 1) return;
 #pragma unroll
-            for (int sit = 0; sit < STRIP; ++sit) {
-                if (!((valid_mask >> sit) & 1)) continue;
-                const int slab = sit * NW + wave;
-                const int xs = slab % NXS;
-                const int cg = (slab / NXS) % CG;
-                const int ry = slab / (NXS * CG);
-                const int xw = xs * 8 + gw;
-                const int x_in0 = xb0 * S + xw * 8;
-                uint4 raw = {0, 0, 0, 0};
-                if ((interior_mask >> sit) & 1) {
-                    if (w_vec) {
-                        raw = __builtin_bit_cast(
-                            uint4, llvm_amdgcn_raw_buffer_load_x4(desc0, (int)voff[sit], 0, 0));
-                        if ((tail_mask >> sit) & 1) {
-                            // zero the x >= W elements of this lane's window
-                            const int rem = p.W - x_in0;  // may be <= 0
-                            uint32_t* dw = reinterpret_cast<uint32_t*>(&raw);
+        for (int sit = 0; sit < STRIP; ++sit) {
+            iregs[sit] = uint4{0, 0, 0, 0};
+            if (!((valid_mask >> sit) & 1)) continue;
+            if ((interior_mask >> sit) & 1 && w_vec) {
+                iregs[sit] = __builtin_bit_cast(
+                    uint4, llvm_amdgcn_raw_buffer_load_x4(desc0, (int)voff[sit], 0, 0));
+                voff[sit] += voff_inc;
+            }
+        }
+    };
+    auto write_input = [&](int cint, char* inb) {
+        if (p.debug == 1) return;
+        const int cin0 = cint * CIN_T;
 #pragma unroll
-                            for (int d = 0; d < 4; ++d) {
-                                const uint32_t m = (rem >= 2 * d + 2)
-                                                       ? 0xffffffffu
-                                                       : ((rem == 2 * d + 1) ? 0x0000ffffu : 0u);
-                                dw[d] &= m;
-                            }
-                        }
-                    } else {
-                        // W % 8 != 0: rows are not 16 B aligned; guarded
-                        // scalar loads (correctness path, non-SD shapes)
-                        const int cin = cin0 + cg * 8 + gr;
-                        const int y_in = yb0 * S - 1 + ry;
-                        if (cin < p.Cin) {
-                            const uint16_t* rp = xin + (int64_t)cin * p.x_sc + (int64_t)y_in * p.W;
-                            uint16_t vals[8];
+        for (int sit = 0; sit < STRIP; ++sit) {
+            if (!((valid_mask >> sit) & 1)) continue;
+            const int slab = sit * NW + wave;
+            const int xs = slab % NXS;
+            const int cg = (slab / NXS) % CG;
+            const int ry = slab / (NXS * CG);
+            const int xw = xs * 8 + gw;
+            const int x_in0 = xb0 * S + xw * 8;
+            uint4 raw = iregs[sit];
+            if ((interior_mask >> sit) & 1) {
+                if (w_vec) {
+                    if ((tail_mask >> sit) & 1) {
+                        // zero the x >= W elements of this lane's window
+                        const int rem = p.W - x_in0;  // may be <= 0
+                        uint32_t* dw = reinterpret_cast<uint32_t*>(&raw);
 #pragma unroll
-                            for (int jj = 0; jj < 8; ++jj)
-                                vals[jj] = (x_in0 + jj < p.W) ? rp[x_in0 + jj] : (uint16_t)0;
-                            raw = *reinterpret_cast<const uint4*>(vals);
+                        for (int d = 0; d < 4; ++d) {
+                            const uint32_t m = (rem >= 2 * d + 2)
+                                                   ? 0xffffffffu
+                                                   : ((rem == 2 * d + 1) ? 0x0000ffffu : 0u);
+                            dw[d] &= m;
                         }
                     }
                 } else {
-                    // halo / zero rows (block at a patch boundary): rare
-                    const int y_in = yb0 * S - 1 + ry;
-                    const uint16_t* src = nullptr;
-                    int64_t sc = 0;
-                    if (y_in == -1 && top) {
-                        src = top;
-                        sc = p.t_sc;
-                    } else if (y_in == p.H && bot) {
-                        src = bot;
-                        sc = p.b_sc;
-                    }
+                    // W % 8 != 0: rows are not 16 B aligned; guarded scalar
+                    // loads (correctness path, non-SD shapes)
                     const int cin = cin0 + cg * 8 + gr;
-                    if (src && cin < p.Cin) {
-                        const uint16_t* rp = src + (int64_t)cin * sc;
+                    const int y_in = yb0 * S - 1 + ry;
+                    if (cin < p.Cin) {
+                        const uint16_t* rp = xin + (int64_t)cin * p.x_sc + (int64_t)y_in * p.W;
                         uint16_t vals[8];
 #pragma unroll
                         for (int jj = 0; jj < 8; ++jj)
@@ -330,38 +316,56 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                         raw = *reinterpret_cast<const uint4*>(vals);
                     }
                 }
-                const uint4 tr = transpose8x8_bf16(raw, lane);
-                const int xi = 1 + xw * 8 + gr;  // lane now owns x = xi
-                if (xi < XIN) {
-                    const int plane = (S == 1) ? 0 : (xi & 1);
-                    const int row = (S == 1) ? xi : (xi >> 1);
-                    char* dst = &inb[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
-                    *reinterpret_cast<uint2*>(dst) = uint2{tr.x, tr.y};
-                    *reinterpret_cast<uint2*>(dst + 8) = uint2{tr.z, tr.w};
-                }
-            }
-#pragma unroll
-            for (int sit = 0; sit < STRIP; ++sit) voff[sit] += voff_inc;
-            // left halo column xi = 0 (x_in = xb0*S - 1): scalar, tiny
-            for (int c = tid; c < YIN * CIN_T; c += NW * WAVE_SIZE) {
-                const int ci = c % CIN_T;
-                const int ry = c / CIN_T;
+            } else {
+                // halo / zero rows (block at a patch boundary): rare
                 const int y_in = yb0 * S - 1 + ry;
-                const int x_in = xb0 * S - 1;
-                const int cin = cin0 + ci;
-                uint16_t val = 0;
-                if (x_in >= 0 && cin < p.Cin) {
-                    if (y_in >= 0 && y_in < p.H)
-                        val = xin[(int64_t)cin * p.x_sc + (int64_t)y_in * p.W + x_in];
-                    else if (y_in == -1 && top)
-                        val = top[(int64_t)cin * p.t_sc + x_in];
-                    else if (y_in == p.H && bot)
-                        val = bot[(int64_t)cin * p.b_sc + x_in];
+                const uint16_t* src = nullptr;
+                int64_t sc = 0;
+                if (y_in == -1 && top) {
+                    src = top;
+                    sc = p.t_sc;
+                } else if (y_in == p.H && bot) {
+                    src = bot;
+                    sc = p.b_sc;
                 }
-                // xi = 0: plane 0, row 0 for both strides
-                *reinterpret_cast<uint16_t*>(
-                    &inb[((0 * YIN + ry) * XP + 0) * ROW_P + ci * 2]) = val;
+                const int cin = cin0 + cg * 8 + gr;
+                if (src && cin < p.Cin) {
+                    const uint16_t* rp = src + (int64_t)cin * sc;
+                    uint16_t vals[8];
+#pragma unroll
+                    for (int jj = 0; jj < 8; ++jj)
+                        vals[jj] = (x_in0 + jj < p.W) ? rp[x_in0 + jj] : (uint16_t)0;
+                    raw = *reinterpret_cast<const uint4*>(vals);
+                }
             }
+            const uint4 tr = transpose8x8_bf16(raw, lane);
+            const int xi = 1 + xw * 8 + gr;  // lane now owns x = xi
+            if (xi < XIN) {
+                const int plane = (S == 1) ? 0 : (xi & 1);
+                const int row = (S == 1) ? xi : (xi >> 1);
+                char* dst = &inb[((plane * YIN + ry) * XP + row) * ROW_P + cg * 16];
+                *reinterpret_cast<uint2*>(dst) = uint2{tr.x, tr.y};
+                *reinterpret_cast<uint2*>(dst + 8) = uint2{tr.z, tr.w};
+            }
+        }
+        // left halo column xi = 0 (x_in = xb0*S - 1): scalar, tiny
+        for (int c = tid; c < YIN * CIN_T; c += NW * WAVE_SIZE) {
+            const int ci = c % CIN_T;
+            const int ry = c / CIN_T;
+            const int y_in = yb0 * S - 1 + ry;
+            const int x_in = xb0 * S - 1;
+            const int cin = cin0 + ci;
+            uint16_t val = 0;
+            if (x_in >= 0 && cin < p.Cin) {
+                if (y_in >= 0 && y_in < p.H)
+                    val = xin[(int64_t)cin * p.x_sc + (int64_t)y_in * p.W + x_in];
+                else if (y_in == -1 && top)
+                    val = top[(int64_t)cin * p.t_sc + x_in];
+                else if (y_in == p.H && bot)
+                    val = bot[(int64_t)cin * p.b_sc + x_in];
+            }
+            // xi = 0: plane 0, row 0 for both strides
+            *reinterpret_cast<uint16_t*>(&inb[((0 * YIN + ry) * XP + 0) * ROW_P + ci * 2]) = val;
         }
     };
     auto compute_tile = [&](char* inb, char* wb) {
@@ -401,16 +405,24 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
         }
     };
 
-    if (n_cin_tiles > 0) stage_tile(0, in_lds, w_lds);
+    if (n_cin_tiles > 0) {
+        stage_weights(0, w_lds);
+        load_input();
+        write_input(0, in_lds);
+    }
     for (int cint = 0; cint < n_cin_tiles; ++cint) {
         char* inb = in_lds + (cint & 1) * IN_SZ;
         char* wb = w_lds + (cint & 1) * W_SZ;
         __syncthreads();  // staged tile cint visible; buffer cint+1 free
-        if (cint + 1 < n_cin_tiles)
-            stage_tile(cint + 1, in_lds + ((cint + 1) & 1) * IN_SZ,
-                       w_lds + ((cint + 1) & 1) * W_SZ);
+        const bool more = cint + 1 < n_cin_tiles;
+        if (more) {
+            stage_weights(cint + 1, w_lds + ((cint + 1) & 1) * W_SZ);
+            load_input();
+        }
         compute_tile(inb, wb);
+        if (more) write_input(cint + 1, in_lds + ((cint + 1) & 1) * IN_SZ);
     }
+
 
     // ---- epilogue: O[b][cout][y][x] = acc + bias ---------------------------
     const int y = yb0 + wy;
