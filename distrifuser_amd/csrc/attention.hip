@@ -69,11 +69,16 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
 template <int NW, int DPAD, int KVB, bool MASK, bool DEFER>
 __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
-    constexpr int VT_ROW = KVB * 2;            // V^T LDS row bytes [d][t]
-    constexpr int K_ROW = pow2ceil(DPAD * 2);  // K LDS row bytes   [t][d]
-    constexpr int KS = DPAD / 16;              // QK^T k-slices
-    constexpr int DT = DPAD / 32;              // PV / O^T d-tiles
-    __shared__ char k_lds[KVB * K_ROW];    // [t][d] bf16, swizzled rows
+    constexpr int VT_ROW = KVB * 2;         // V^T LDS row bytes [d][t]
+    // K rows padded by 8 B: row stride DPAD*2+8 gives gcd(stride/4, 32) = 2,
+    // so the 32-consecutive-row column reads are 2-way (free) instead of the
+    // 4-way floor of XOR-swizzled 128 B rows — same trick as the conv
+    // kernel's staged tile (profiles/conv_ladder_r02.md v5). Reads/writes
+    // are b64 pairs (rows are 8 B- but not 16 B-aligned).
+    constexpr int K_ROW = DPAD * 2 + 8;     // padded K LDS row bytes [t][d]
+    constexpr int KS = DPAD / 16;           // QK^T k-slices
+    constexpr int DT = DPAD / 32;           // PV / O^T d-tiles
+    __shared__ char k_lds[KVB * K_ROW];     // [t][d] bf16, padded rows
     __shared__ char vt_lds[DPAD * VT_ROW];  // [d][t] bf16, swizzled rows
 
     const int tid = threadIdx.x;
@@ -149,8 +154,9 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
                     kraw = *reinterpret_cast<const uint4*>(
                         kbase + chunk * p.k_sc + tin * p.k_sl + d8 * 8);
                 }
-                *reinterpret_cast<uint4*>(
-                    &k_lds[t_local * K_ROW + vt_swz<K_ROW>(t_local, d8 * 16)]) = kraw;
+                char* kdst = &k_lds[t_local * K_ROW + d8 * 16];
+                *reinterpret_cast<uint2*>(kdst) = uint2{kraw.x, kraw.y};
+                *reinterpret_cast<uint2*>(kdst + 8) = uint2{kraw.z, kraw.w};
             }
             // V^T: wave w stages token rows [w*8, w*8+8) of each 8*NW-row pass
             constexpr int VROWS_PER_PASS = NW * 8;
@@ -186,8 +192,11 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
 #pragma unroll
             for (int ks = 0; ks < KS; ++ks) {
                 const int t = st * 32 + lo;
-                short8 kfrag = *reinterpret_cast<const short8*>(
-                    &k_lds[t * K_ROW + vt_swz<K_ROW>(t, (ks * 16 + hi * 8) * 2)]);
+                const char* ksrc = &k_lds[t * K_ROW + (ks * 16 + hi * 8) * 2];
+                const uint2 k0 = *reinterpret_cast<const uint2*>(ksrc);
+                const uint2 k1 = *reinterpret_cast<const uint2*>(ksrc + 8);
+                const uint4 kk{k0.x, k0.y, k1.x, k1.y};
+                const short8 kfrag = __builtin_bit_cast(short8, kk);
                 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
