@@ -70,6 +70,72 @@ __device__ __forceinline__ float geluf(float x) {
     return 0.5f * x * (1.f + erff(x * 0.70710678118654752440f));
 }
 
+// In-register 8x8 bf16 transpose across an 8-lane group (butterfly): group
+// member r = (lane&3)|((lane&8)>>1) holds row r (8 elements = one uint4);
+// afterwards member r holds column r. Lane-mask set {1,2,8} is chosen so
+// every exchange is a VALU DPP op (quad_perm for ^1/^2, row_ror:8 for ^8)
+// — __shfl_xor compiles to ds_bpermute, which rides the SAME LDS pipe the
+// MFMA fragment reads need (v3 PMC: 2.9 LDS instrs/MFMA, 19% MFMA util).
+// Stage (lane-bit m, elem-bit e): where bit(lane,m) != bit(elem,e):
+// new[c] = partner(lane^m)'s e[c^e]; composition over the three stages is
+// the full transpose (verified by simulation).
+__device__ __forceinline__ uint32_t dpp_xor1(uint32_t v) {
+    return __builtin_amdgcn_mov_dpp(v, 0xB1, 0xf, 0xf, true);  // quad_perm [1,0,3,2]
+}
+__device__ __forceinline__ uint32_t dpp_xor2(uint32_t v) {
+    return __builtin_amdgcn_mov_dpp(v, 0x4E, 0xf, 0xf, true);  // quad_perm [2,3,0,1]
+}
+__device__ __forceinline__ uint32_t dpp_xor8(uint32_t v) {
+    return __builtin_amdgcn_mov_dpp(v, 0x128, 0xf, 0xf, true);  // row_ror:8
+}
+
+__device__ __forceinline__ uint4 transpose8x8_bf16(uint4 v, int lane) {
+    uint32_t d[4] = {v.x, v.y, v.z, v.w};
+    // stage (m=1, e=1): bf16 halves within dwords
+    {
+        const bool hi = (lane & 1) != 0;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const uint32_t sw = (d[i] >> 16) | (d[i] << 16);  // e[c^1]
+            const uint32_t recv = dpp_xor1(sw);
+            d[i] = hi ? (d[i] & 0xffff0000u) | (recv & 0x0000ffffu)
+                      : (d[i] & 0x0000ffffu) | (recv & 0xffff0000u);
+        }
+    }
+    // stage (m=2, e=2): dword pairs (0<->1, 2<->3)
+    {
+        const bool hi = (lane & 2) != 0;
+        const uint32_t r0 = dpp_xor2(d[1]);
+        const uint32_t r1 = dpp_xor2(d[0]);
+        const uint32_t r2 = dpp_xor2(d[3]);
+        const uint32_t r3 = dpp_xor2(d[2]);
+        if (hi) {
+            d[0] = r0;
+            d[2] = r2;
+        } else {
+            d[1] = r1;
+            d[3] = r3;
+        }
+    }
+    // stage (m=8, e=4): dword pairs (0<->2, 1<->3)
+    {
+        const bool hi = (lane & 8) != 0;
+        const uint32_t r0 = dpp_xor8(d[2]);
+        const uint32_t r1 = dpp_xor8(d[3]);
+        const uint32_t r2 = dpp_xor8(d[0]);
+        const uint32_t r3 = dpp_xor8(d[1]);
+        if (hi) {
+            d[0] = r0;
+            d[1] = r1;
+        } else {
+            d[2] = r2;
+            d[3] = r3;
+        }
+    }
+    return uint4{d[0], d[1], d[2], d[3]};
+}
+
+
 #define DFA_HIP_CHECK(expr)                                                     \
     do {                                                                        \
         hipError_t _e = (expr);                                                 \
