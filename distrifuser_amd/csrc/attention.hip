@@ -78,8 +78,8 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
     constexpr int K_ROW = DPAD * 2 + 8;     // padded K LDS row bytes [t][d]
     constexpr int KS = DPAD / 16;           // QK^T k-slices
     constexpr int DT = DPAD / 32;           // PV / O^T d-tiles
-    __shared__ char k_lds[2 * KVB * K_ROW];     // [t][d] bf16, padded rows
-    __shared__ char vt_lds[2 * DPAD * VT_ROW];  // [d][t] bf16, swizzled rows
+    __shared__ char k_lds[KVB * K_ROW];     // [t][d] bf16, padded rows
+    __shared__ char vt_lds[DPAD * VT_ROW];  // [d][t] bf16, swizzled rows
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
@@ -212,15 +212,18 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
         }
     };
 
+    // Single-buffered (double buffers halved blocks/CU and LOST 13% at
+    // L=57.6k): tile t+1's loads are issued before tile t's compute and
+    // land under it; the write phase between the two barriers is short.
     if (n_tiles > 0) {
         load_tile(0);
         write_tile(k_lds, vt_lds);
     }
     for (int tile = 0; tile < n_tiles; ++tile) {
-        char* kb = k_lds + (tile & 1) * (KVB * K_ROW);
-        char* vb = vt_lds + (tile & 1) * (DPAD * VT_ROW);
+        char* kb = k_lds;
+        char* vb = vt_lds;
         const int64_t t0 = (int64_t)tile * KVB;
-        __syncthreads();  // staged tile visible; other buffer free
+        __syncthreads();  // staged tile visible
         const bool more = tile + 1 < n_tiles;
         if (more) load_tile(tile + 1);
 
@@ -318,9 +321,8 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
             __builtin_amdgcn_s_setprio(0);
         }
 
-        if (more)
-            write_tile(k_lds + ((tile + 1) & 1) * (KVB * K_ROW),
-                       vt_lds + ((tile + 1) & 1) * (DPAD * VT_ROW));
+        __syncthreads();  // all waves done reading tile `tile`
+        if (more) write_tile(k_lds, vt_lds);
     }
 
     // ---- epilogue: O[q][d] = O^T / l (only the real head_dim columns) ----
@@ -351,21 +353,12 @@ static void launch_var(const FlashAttnParams& p, hipStream_t stream) {
 template <int DPAD>
 static void launch_dpad(const FlashAttnParams& p, hipStream_t stream) {
     const int64_t Lkv = p.NC * p.LC;
-    if constexpr (DPAD >= 160) {
-        // double-buffered 128-token tiles exceed the 160 KB LDS at d=160;
-        // 64-token tiles fit (SD1.5's 160-dim heads only)
-        if (Lkv % 64 == 0)
-            launch_var<8, DPAD, 64, false>(p, stream);
-        else
-            launch_var<8, DPAD, 64, true>(p, stream);
-    } else {
-        if (Lkv % 128 == 0)
-            launch_var<8, DPAD, 128, false>(p, stream);
-        else if (Lkv % 64 == 0)
-            launch_var<8, DPAD, 64, false>(p, stream);
-        else
-            launch_var<8, DPAD, 128, true>(p, stream);
-    }
+    if (Lkv % 128 == 0)
+        launch_var<8, DPAD, 128, false>(p, stream);
+    else if (Lkv % 64 == 0)
+        launch_var<8, DPAD, 64, false>(p, stream);
+    else
+        launch_var<8, DPAD, 128, true>(p, stream);
 }
 
 void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream) {
