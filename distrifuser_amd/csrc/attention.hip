@@ -215,7 +215,8 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
     // Single-buffered (double buffers halved blocks/CU and LOST 13% at
     // L=57.6k): tile t+1's loads are issued before tile t's compute and
     // land under it; the write phase between the two barriers is short.
-    if (n_tiles > 0) {
+    const bool pipe = p.pipe_mode == 0;
+    if (pipe && n_tiles > 0) {
         load_tile(0);
         write_tile(k_lds, vt_lds);
     }
@@ -223,8 +224,12 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
         char* kb = k_lds;
         char* vb = vt_lds;
         const int64_t t0 = (int64_t)tile * KVB;
+        if (!pipe) {
+            load_tile(tile);
+            write_tile(k_lds, vt_lds);
+        }
         __syncthreads();  // staged tile visible
-        const bool more = tile + 1 < n_tiles;
+        const bool more = pipe && tile + 1 < n_tiles;
         if (more) load_tile(tile + 1);
 
 #pragma unroll
