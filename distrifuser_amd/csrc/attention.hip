@@ -129,108 +129,60 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
     const float defer_raw = 11.0f / scale2;
 
     const int n_tiles = (int)((Lkv + KVB - 1) / KVB);
-
-    // ---- double-buffered KV staging (the conv-kernel pipeline pattern,
-    // profiles/conv_ladder_r02.md v8/v9): loads for tile t+1 are issued
-    // right after the barrier into registers, land under tile t's compute,
-    // and are written to the other LDS buffer afterwards (__syncthreads
-    // drains vmcnt, so in-flight loads cannot cross a barrier).
-    // V is loaded as row-major b128 bursts and flipped to the [d][t] layout
-    // with the DPP 8x8 butterfly — the previous column-load staging issued
-    // 8 scalar 2 B loads per item.
-    constexpr int KCOLS = DPAD / 8;  // 16 B chunks per K row
-    constexpr int KITEMS = (KVB * KCOLS + NW * WAVE_SIZE - 1) / (NW * WAVE_SIZE);
-    constexpr int VPASS = KVB / (NW * 8);      // 8-token windows per wave
-    constexpr int VDBLK = (DPAD + 63) / 64;    // 64-d blocks per pass
-    const int gr = (lane & 3) | ((lane & 8) >> 1);   // transpose-group member
-    const int gw = ((lane >> 2) & 1) | (((lane >> 4) & 3) << 1);
-    uint4 kreg[KITEMS];
-    uint4 vreg[VPASS * VDBLK];
-
-    auto load_tile = [&](int tile) {
+    for (int tile = 0; tile < n_tiles; ++tile) {
         const int64_t t0 = (int64_t)tile * KVB;
+        // ---- cooperative staging ----
+        // K [t][d]: thread (t-row, 16B column) -> b128 LDS write.
+        // V^T [d][t]: lane = d COLUMN, 8 consecutive tokens per pass — the
+        // global reads stay perfectly coalesced (64 lanes x 2B = one 128B row
+        // per token) and the LDS write becomes ONE b128 per lane instead of
+        // 8 scalar ds_write_b16 (the transpose-write conflicts were 15% of
+        // kernel time — see profiles/attention_ladder_r01.md ablation).
+        {
+            constexpr int KCOLS = DPAD / 8;  // 16 B chunks per K row
+            constexpr int NCHUNK = KVB * KCOLS;
 #pragma unroll
-        for (int it = 0; it < KITEMS; ++it) {
-            const int c = it * NW * WAVE_SIZE + tid;
-            kreg[it] = uint4{0, 0, 0, 0};
-            if (KVB * KCOLS % (NW * WAVE_SIZE) == 0 || c < KVB * KCOLS) {
+            for (int c = tid; c < NCHUNK; c += NW * WAVE_SIZE) {
                 const int t_local = c / KCOLS;
                 const int d8 = c % KCOLS;
                 const int64_t t_glob = t0 + t_local;
+                uint4 kraw = {0, 0, 0, 0};
                 const bool d_ok = (p.Dh == DPAD) || (d8 * 8 + 8 <= p.Dh);
                 if (d_ok && (!MASK || t_glob < Lkv)) {
                     const int64_t chunk = t_glob / p.LC;
                     const int64_t tin = t_glob % p.LC;
-                    kreg[it] = *reinterpret_cast<const uint4*>(
+                    kraw = *reinterpret_cast<const uint4*>(
                         kbase + chunk * p.k_sc + tin * p.k_sl + d8 * 8);
+                }
+                char* kdst = &k_lds[t_local * K_ROW + d8 * 16];
+                *reinterpret_cast<uint2*>(kdst) = uint2{kraw.x, kraw.y};
+                *reinterpret_cast<uint2*>(kdst + 8) = uint2{kraw.z, kraw.w};
+            }
+            // V^T: wave w stages token rows [w*8, w*8+8) of each 8*NW-row pass
+            constexpr int VROWS_PER_PASS = NW * 8;
+#pragma unroll
+            for (int rep = 0; rep < KVB / VROWS_PER_PASS; ++rep) {
+                const int tb_local = rep * VROWS_PER_PASS + wave * 8;
+                const int64_t tb_glob = t0 + tb_local;
+                // tb_glob is a multiple of 8 and LC % 8 == 0 (host-checked),
+                // so the 8-token window lies in one chunk
+                const int64_t chunk = tb_glob / p.LC;
+                const int64_t tin = tb_glob % p.LC;
+                for (int d = lane; d < DPAD; d += WAVE_SIZE) {
+                    uint16_t ve[8];
+                    const uint16_t* vp = vbase + chunk * p.v_sc + tin * p.v_sl + d;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const bool ok = d < p.Dh && (!MASK || tb_glob + j < Lkv);
+                        ve[j] = ok ? vp[j * p.v_sl] : (uint16_t)0;
+                    }
+                    *reinterpret_cast<uint4*>(
+                        &vt_lds[d * VT_ROW + vt_swz<VT_ROW>(d, tb_local * 2)]) =
+                        *reinterpret_cast<const uint4*>(ve);
                 }
             }
         }
-#pragma unroll
-        for (int rep = 0; rep < VPASS; ++rep) {
-            const int tb_local = (rep * NW + wave) * 8;
-            const int64_t tb_glob = t0 + tb_local;
-            // tb_glob is a multiple of 8 and LC % 8 == 0 (host-checked),
-            // so the 8-token window lies in one chunk
-            const int64_t chunk = tb_glob / p.LC;
-            const int64_t tin = tb_glob % p.LC;
-#pragma unroll
-            for (int db = 0; db < VDBLK; ++db) {
-                const int d0 = (db * 8 + gw) * 8;
-                uint4 raw = {0, 0, 0, 0};
-                const bool d_ok = d0 < DPAD && ((p.Dh == DPAD) || (d0 + 8 <= p.Dh));
-                if (d_ok && (!MASK || tb_glob + gr < Lkv))
-                    raw = *reinterpret_cast<const uint4*>(
-                        vbase + chunk * p.v_sc + (tin + gr) * p.v_sl + d0);
-                vreg[rep * VDBLK + db] = raw;
-            }
-        }
-    };
-    auto write_tile = [&](char* kb, char* vb) {
-#pragma unroll
-        for (int it = 0; it < KITEMS; ++it) {
-            const int c = it * NW * WAVE_SIZE + tid;
-            if (KVB * KCOLS % (NW * WAVE_SIZE) == 0 || c < KVB * KCOLS) {
-                const int t_local = c / KCOLS;
-                const int d8 = c % KCOLS;
-                char* kdst = &kb[t_local * K_ROW + d8 * 16];
-                *reinterpret_cast<uint2*>(kdst) = uint2{kreg[it].x, kreg[it].y};
-                *reinterpret_cast<uint2*>(kdst + 8) = uint2{kreg[it].z, kreg[it].w};
-            }
-        }
-#pragma unroll
-        for (int rep = 0; rep < VPASS; ++rep) {
-            const int tb_local = (rep * NW + wave) * 8;
-#pragma unroll
-            for (int db = 0; db < VDBLK; ++db) {
-                const uint4 tr = transpose8x8_bf16(vreg[rep * VDBLK + db], lane);
-                const int d = (db * 8 + gw) * 8 + gr;  // lane owns this d row
-                if (d < DPAD)
-                    *reinterpret_cast<uint4*>(
-                        &vb[d * VT_ROW + vt_swz<VT_ROW>(d, tb_local * 2)]) = tr;
-            }
-        }
-    };
-
-    // Single-buffered (double buffers halved blocks/CU and LOST 13% at
-    // L=57.6k): tile t+1's loads are issued before tile t's compute and
-    // land under it; the write phase between the two barriers is short.
-    const bool pipe = p.pipe_mode == 0;
-    if (pipe && n_tiles > 0) {
-        load_tile(0);
-        write_tile(k_lds, vt_lds);
-    }
-    for (int tile = 0; tile < n_tiles; ++tile) {
-        char* kb = k_lds;
-        char* vb = vt_lds;
-        const int64_t t0 = (int64_t)tile * KVB;
-        if (!pipe) {
-            load_tile(tile);
-            write_tile(k_lds, vt_lds);
-        }
-        __syncthreads();  // staged tile visible
-        const bool more = pipe && tile + 1 < n_tiles;
-        if (more) load_tile(tile + 1);
+        __syncthreads();
 
 #pragma unroll
         for (int st = 0; st < KVB / 32; ++st) {  // 32-token sub-tiles
@@ -240,7 +192,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
 #pragma unroll
             for (int ks = 0; ks < KS; ++ks) {
                 const int t = st * 32 + lo;
-                const char* ksrc = &kb[t * K_ROW + (ks * 16 + hi * 8) * 2];
+                const char* ksrc = &k_lds[t * K_ROW + (ks * 16 + hi * 8) * 2];
                 const uint2 k0 = *reinterpret_cast<const uint2*>(ksrc);
                 const uint2 k1 = *reinterpret_cast<const uint2*>(ksrc + 8);
                 const uint4 kk{k0.x, k0.y, k1.x, k1.y};
@@ -319,15 +271,13 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
                 for (int kt = 0; kt < 2; ++kt) {
                     const int d = dt * 32 + lo;
                     short8 vf = *reinterpret_cast<const short8*>(
-                        &vb[d * VT_ROW + vt_swz<VT_ROW>(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
+                        &vt_lds[d * VT_ROW + vt_swz<VT_ROW>(d, (st * 32 + kt * 16 + hi * 8) * 2)]);
                     ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[kt], ot[dt], 0, 0, 0);
                 }
             }
             __builtin_amdgcn_s_setprio(0);
         }
-
-        __syncthreads();  // all waves done reading tile `tile`
-        if (more) write_tile(k_lds, vt_lds);
+        __syncthreads();
     }
 
     // ---- epilogue: O[q][d] = O^T / l (only the real head_dim columns) ----

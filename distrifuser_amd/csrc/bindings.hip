@@ -164,7 +164,6 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
     p.q_sl = q.stride(2);
     p.Dh = Dh;
     p.scale = 1.0f / std::sqrt((float)Dh);
-    if (const char* m = getenv("DFA_ATTN_NOPIPE")) p.pipe_mode = atoi(m);
 
     auto set_kv = [&](const at::Tensor& t, const uint16_t*& ptr, int64_t& sb, int64_t& sh,
                       int64_t& sc, int64_t& sl, int64_t& NC, int64_t& LC) {

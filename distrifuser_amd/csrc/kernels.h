@@ -81,6 +81,5 @@ struct FlashAttnParams {
     int64_t k_sb, k_sh, k_sc, k_sl;
     int64_t v_sb, v_sh, v_sc, v_sl;
     float scale;  // 1/sqrt(64)
-    int pipe_mode;  // 0 = pipelined staging (default); 1 = stage-then-compute (A/B)
 };
 void launch_flash_attention_d64(const FlashAttnParams& p, hipStream_t stream);
