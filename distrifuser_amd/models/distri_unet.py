@@ -99,20 +99,29 @@ class DistriUNet(nn.Module):
         local = local.contiguous()
         numel = local.numel()
         if self.buffer_list is None or self.buffer_list[0].numel() != numel:
-            flat = torch.empty(ws, numel, device=local.device, dtype=local.dtype)
-            self.buffer_list = [flat[i] for i in range(ws)]
+            self._gather_flat = torch.empty(ws, numel, device=local.device, dtype=local.dtype)
+            self.buffer_list = [self._gather_flat[i] for i in range(ws)]
             self.output_buffer = None
         dist.all_gather(self.buffer_list, local.reshape(-1), async_op=False)
-        views = [b.view(local.shape) for b in self.buffer_list]
-        if cfg.split_batch:
-            b0 = torch.cat(views[:n], dim=dim)
-            b1 = torch.cat(views[n:], dim=dim)
-            out = torch.cat([b0, b1], dim=0)
+        # Reassemble with ONE strided copy into the stable output buffer
+        # (SURVEY K11): the reference materializes torch.cat per step
+        # (/root/reference/distrifuser/models/distri_sdxl_unet_pp.py:166-168)
+        # and the round-1 code paid cat + copy_ — two full passes over
+        # (2,C,H,W) where one suffices.
+        lb, c, hl, wl = local.shape
+        groups = 2 if cfg.split_batch else 1
+        per = ws // groups
+        out_shape = [groups * lb, c, hl, wl]
+        out_shape[dim] = local.shape[dim] * per
+        if self.output_buffer is None or list(self.output_buffer.shape) != out_shape:
+            self.output_buffer = torch.empty(out_shape, device=local.device, dtype=local.dtype)
+        src = self._gather_flat.view(groups, per, lb, c, hl, wl)
+        if dim == 2:
+            dst = self.output_buffer.view(groups, lb, c, per, hl, wl)
+            dst.copy_(src.permute(0, 2, 3, 1, 4, 5))
         else:
-            out = torch.cat(views, dim=dim)
-        if self.output_buffer is None or self.output_buffer.shape != out.shape:
-            self.output_buffer = torch.empty_like(out)
-        self.output_buffer.copy_(out)
+            dst = self.output_buffer.view(groups, lb, c, hl, per, wl)
+            dst.copy_(src.permute(0, 2, 3, 4, 1, 5))
         return self.output_buffer
 
     def _slice_cfg(self, sample, timestep, encoder_hidden_states, added_cond_kwargs):
