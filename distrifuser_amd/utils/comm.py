@@ -49,8 +49,19 @@ class PatchParallelismCommManager:
         # batching state
         self.idx_queue: list[int] = []
 
-        # observability: per-collective accounting (bytes pushed per rank)
-        self.stats = {"gathers": 0, "bytes": 0}
+        # observability: per-collective accounting (bytes pushed per rank);
+        # DFA_COMM_TIMING=1 adds per-gather device timing via event pairs
+        # (skipped inside hipGraph capture, where events are not allowed)
+        self.stats = {"gathers": 0, "bytes": 0, "gather_ms": []}
+        self._timing = os.environ.get("DFA_COMM_TIMING", "0") == "1"
+        # DFA_COMM_SIDE_STREAM=1 issues the batched gather from a dedicated
+        # HIP stream with event handoff (SURVEY §2.4b xGMI mapping). RCCL
+        # already runs collectives on its own internal stream, so this
+        # mainly decouples the gather ENQUEUE from the compute stream; kept
+        # opt-in until a multi-GPU measurement shows a win.
+        self._use_side = os.environ.get("DFA_COMM_SIDE_STREAM", "0") == "1"
+        self._side_stream: torch.cuda.Stream | None = None
+        self._timing_events: list = []
 
     # -- registration pass -------------------------------------------------
 
@@ -151,10 +162,35 @@ class PatchParallelismCommManager:
         group = self.distri_config.batch_group
         own = self.distri_config.split_idx()
         tensor_list = [self.buffer[p, start:end] for p in range(self.buffer.shape[0])]
-        handle = dist.all_gather(
-            tensor_list, self.buffer[own, start:end], group=group,
-            async_op=not DEBUG_SYNC,
-        )
+        on_gpu = self.buffer.is_cuda
+        capturing = on_gpu and torch.cuda.is_current_stream_capturing()
+        time_this = self._timing and on_gpu and not capturing
+        if time_this:
+            ev0 = torch.cuda.Event(enable_timing=True)
+            ev1 = torch.cuda.Event(enable_timing=True)
+
+        def _issue():
+            if time_this:
+                ev0.record()
+            h = dist.all_gather(
+                tensor_list, self.buffer[own, start:end], group=group,
+                async_op=not DEBUG_SYNC,
+            )
+            if time_this:
+                ev1.record()
+                self._timing_events.append((ev0, ev1))
+            return h
+
+        if self._use_side and on_gpu and not capturing:
+            if self._side_stream is None:
+                self._side_stream = torch.cuda.Stream()
+            ready = torch.cuda.Event()
+            ready.record()  # buffer slot writes on the compute stream
+            with torch.cuda.stream(self._side_stream):
+                self._side_stream.wait_event(ready)
+                handle = _issue()
+        else:
+            handle = _issue()
         self.stats["gathers"] += 1
         self.stats["bytes"] += (end - start) * self.buffer.element_size()
         for idx in self.idx_queue:
@@ -179,3 +215,13 @@ class PatchParallelismCommManager:
             if h is not None:
                 h.wait()
                 self.handles[i] = None
+        self.harvest_timing()
+
+    def harvest_timing(self) -> None:
+        """Fold completed per-gather event pairs into stats["gather_ms"]."""
+        if not self._timing_events:
+            return
+        torch.cuda.synchronize()
+        for ev0, ev1 in self._timing_events:
+            self.stats["gather_ms"].append(ev0.elapsed_time(ev1))
+        self._timing_events = []
