@@ -258,6 +258,26 @@ at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
     return o;
 }
 
+at::Tensor vae_attention(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v) {
+    TORCH_CHECK(q.is_cuda() && q.dim() == 3 && q.size(-1) == 512 &&
+                q.scalar_type() == at::kBFloat16,
+                "vae_attention: bf16 [B, L, 512] only");
+    auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+    TORCH_CHECK(kc.sizes() == qc.sizes() && vc.sizes() == qc.sizes());
+    VaeAttnParams p{};
+    p.q = reinterpret_cast<const uint16_t*>(qc.data_ptr());
+    p.k = reinterpret_cast<const uint16_t*>(kc.data_ptr());
+    p.v = reinterpret_cast<const uint16_t*>(vc.data_ptr());
+    p.B = (int)qc.size(0);
+    p.L = qc.size(1);
+    p.sb = qc.stride(0);
+    p.scale = 1.0f / std::sqrt(512.0f);
+    auto o = at::empty_like(qc);
+    p.o = reinterpret_cast<uint16_t*>(o.data_ptr());
+    launch_vae_attention(p, cur_stream());
+    return o;
+}
+
 std::vector<at::Tensor> mfma_probe(const at::Tensor& a, const at::Tensor& b);
 
 }  // namespace
@@ -295,6 +315,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "merge stale peer GN moments with fresh local ones (one launch)");
     m.def("cfg_affine_step", &cfg_affine_step, "fused CFG combine + affine scheduler update");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
+    m.def("vae_attention", &vae_attention, "bf16 single-head d=512 VAE mid attention");
     m.def("conv3x3", &conv3x3,
           "bf16 implicit-GEMM 3x3 conv, stride 1/2, in-place halo rows");
     m.def("mfma_probe", &mfma_probe, "dump mfma_f32_16x16x32_bf16 fragment mapping");

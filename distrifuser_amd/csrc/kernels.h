@@ -65,6 +65,19 @@ struct Conv3x3Params {
 };
 void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream);
 
+// ---- VAE mid-block attention (bf16, single head, head_dim 512) --------------
+// q/k/v/o: [B][L][512] row-contiguous (sb = batch stride in elements).
+struct VaeAttnParams {
+    const uint16_t* q;
+    const uint16_t* k;
+    const uint16_t* v;
+    uint16_t* o;
+    int B;
+    int64_t L, sb;
+    float scale;
+};
+void launch_vae_attention(const VaeAttnParams& p, hipStream_t stream);
+
 // ---- Flash attention (bf16, SD-family head dims) ----------------------------
 // q: logical [B, H, Lq, 64]; k/v: logical [B, H, NC, LC, 64] (NC stale-KV
 // chunks of LC tokens; NC=1 for plain attention). All strides in ELEMENTS,

@@ -3,9 +3,11 @@
 The reference delegated VAE decode to diffusers; here it is owned. Every
 rank decodes the identical full latent (parity with the reference's
 replicated decode — the denoise loop is the parallel part). The mid-block
-attention is evaluated with a query-chunked online-softmax fallback so that
-3840x3840 (480x480 = 230k tokens, single 512-dim head) decodes without
-materializing the score matrix; on GPU it rides the fused SDPA.
+attention (single 512-dim head over up to 480x480 = 230k tokens at
+3840x3840) runs on the split-D gfx950 HIP kernel (csrc/vae_attn.hip) on
+GPU; the CPU path uses a query-chunked softmax so no full score matrix is
+ever materialized. Convolutions ride the implicit-GEMM conv3x3 kernel
+(NativeConv2d).
 """
 
 from __future__ import annotations
@@ -104,7 +106,9 @@ class VAEAttention(nn.Module):
                                 self.group_norm.bias, self.group_norm.eps, silu=False)
         x = x.permute(0, 2, 3, 1).reshape(b, h * w, c)
         q, k, v = self.to_q(x), self.to_k(x), self.to_v(x)
-        if x.is_cuda and c <= 256:
+        if x.is_cuda and c == 512 and q.dtype == torch.bfloat16:
+            out = ops.vae_attention(q, k, v)  # split-D HIP kernel
+        elif x.is_cuda and c <= 256:
             out = ops.flash_attention(q[:, None], k[:, None], v[:, None])[:, 0]
         else:
             out = _chunked_single_head_attention(q, k, v)

@@ -16,6 +16,7 @@ from .dispatch import (
     group_norm_silu,
     group_norm_stats,
     hip_ext,
+    vae_attention,
     hip_ext_available,
 )
 
@@ -31,4 +32,5 @@ __all__ = [
     "hip_ext",
     "hip_ext_available",
     "pack_conv3x3_weight",
+    "vae_attention",
 ]

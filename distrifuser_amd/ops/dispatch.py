@@ -135,6 +135,13 @@ def group_norm_silu(x, num_groups, weight, bias, eps, silu=True):
     return eager.group_norm_silu(x, num_groups, weight, bias, eps, silu)
 
 
+def vae_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """Single-head d=512 attention, q/k/v [B, L, 512] (VAE mid block)."""
+    if _use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] == 512:
+        return hip_ext().vae_attention(q, k, v)
+    return eager.vae_attention(q, k, v)
+
+
 def geglu(hidden: torch.Tensor) -> torch.Tensor:
     if _use_hip(hidden):
         return hip_ext().geglu(hidden)

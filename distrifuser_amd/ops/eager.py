@@ -89,6 +89,14 @@ def geglu(hidden: torch.Tensor) -> torch.Tensor:
     return a * F.gelu(b)
 
 
+def vae_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """Single-head attention over [B, L, C] (fp32 softmax)."""
+    scale = q.shape[-1] ** -0.5
+    scores = torch.einsum("bqc,bkc->bqk", q.float() * scale, k.float())
+    probs = scores.softmax(dim=-1)
+    return torch.einsum("bqk,bkc->bqc", probs, v.float()).to(q.dtype)
+
+
 def conv3x3_halo(
     x: torch.Tensor,
     weight: torch.Tensor,

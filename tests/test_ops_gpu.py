@@ -360,3 +360,22 @@ def test_native_conv2d_module_matches_eager():
         x.float(), m.weight.float(), m.bias.float(), padding=1)
     err = (got.float() - ref).abs().max().item()
     assert err <= 0.02 * max(ref.abs().max().item(), 1.0)
+
+
+# ---- VAE mid-attention (single head, d=512) --------------------------------
+
+
+@requires_gpu
+@pytest.mark.parametrize("b,l", [(1, 256), (2, 1000), (1, 4096)])
+def test_vae_attention_vs_fp32(b, l):
+    from distrifuser_amd import ops
+    from distrifuser_amd.ops import eager
+
+    torch.manual_seed(0)
+    q = torch.randn(b, l, 512, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, l, 512, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, l, 512, device="cuda", dtype=torch.bfloat16)
+    got = ops.hip_ext().vae_attention(q, k, v)
+    ref = eager.vae_attention(q.float(), k.float(), v.float())
+    err = (got.float() - ref).abs().max().item()
+    assert err <= 0.03, f"max err {err}"
