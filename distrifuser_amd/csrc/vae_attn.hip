@@ -4,25 +4,23 @@
 // d=512 cannot ride the d64 flash kernel: per-lane O accumulators would be
 // 16 MFMA tiles (256 VGPRs) and Q fragments 32 k-slices (256 VGPRs).
 // Split-D design instead:
-// * one 64-query block; 8 waves each own a 64-wide d-slice (Q fragments
-//   2x32 VGPRs, O accumulators 2x32 per wave for the two 32-query tiles);
+// * one 32-query tile per block; 4 waves each own a 128-wide d-slice
+//   (Q fragments 64 VGPRs, O accumulators 64 VGPRs per wave);
 // * per 32-token KV tile, each wave computes its PARTIAL S^T (its d-slice's
-//   QK contribution, swapped mfma_f32_32x32x16_bf16 like the d64 kernel)
-//   and ds_add_f32-accumulates it into one shared S tile; the reduction
-//   pass applies the shared online-softmax state (m, l per query, in LDS)
-//   and materializes P[q][t] bf16;
+//   QK contribution, swapped mfma_f32_32x32x16_bf16 like the d64 kernel),
+//   written to a per-wave LDS slab; a split-K reduction sums the slabs,
+//   updates the shared online-softmax state (m, l per query, in LDS), and
+//   materializes P[q][t] bf16;
 // * each wave then rescales its O^T accumulators by the shared correction
 //   and accumulates PV for its own d-slice;
-// * the epilogue transposes O through LDS (aliasing the K/V buffers) so
-//   the global stores are row-contiguous ([L][512] layout).
+// * the epilogue transposes O through LDS so the global stores are
+//   row-contiguous ([L][512] layout).
 // LDS rows are 8 B-padded (stride 1032/72 B: gcd(dwords,32)=2 -> 2-way
 // conflicts) and fragment reads are b64 pairs, as in the conv kernel
 // (profiles/conv_ladder_r02.md v5).
 //
 // Replaces the chunked bf16 einsum fallback in models/vae.py, which
-// materialized fp32 score slabs. v1 of this kernel
-// used 4 waves x 32 queries: 64 TF at L=16k, bound by KV re-stream; this
-// 64-query version doubles the compute per staged KV byte.
+// materialized fp32 score slabs.
 
 #include "common.h"
 #include "kernels.h"
@@ -31,23 +29,18 @@ namespace {
 
 typedef float float16v __attribute__((ext_vector_type(16)));
 
-constexpr int NW = 8;        // waves = d-slices
-constexpr int QT = 64;       // queries per block (two 32-row MFMA tiles)
-constexpr int NQT = QT / 32;
+constexpr int NW = 4;        // waves = d-slices
+constexpr int QT = 32;       // queries per block
 constexpr int KVB = 32;      // kv tokens per tile
 constexpr int C = 512;       // channels (head_dim)
-constexpr int DSL = C / NW;  // d-slice per wave (64)
+constexpr int DSL = C / NW;  // d-slice per wave (128)
 constexpr int KS = DSL / 16;
 constexpr int DT = DSL / 32;
 
-constexpr int K_ROW = C * 2 + 8;     // [t][d] row bytes
-constexpr int VT_ROW = KVB * 2 + 8;  // [d][t] row bytes
-constexpr int P_ROW = KVB * 2 + 8;   // [q][t] row bytes
-constexpr int O_ROW = C * 2 + 8;     // [q][d] row bytes (aliases k+vt)
-
-constexpr int K_SZ = KVB * K_ROW;
-constexpr int VT_SZ = C * VT_ROW;
-constexpr int P_SZ = QT * P_ROW;
+constexpr int K_ROW = C * 2 + 8;    // k_lds row bytes [t][d]
+constexpr int VT_ROW = KVB * 2 + 8; // vt_lds row bytes [d][t]
+constexpr int P_ROW = KVB * 2 + 8;  // p_lds row bytes [q][t]
+constexpr int O_ROW = C * 2 + 8;    // o_lds row bytes [q][d] (aliases k_lds)
 
 __device__ __forceinline__ short8 lds_frag_b64x2(const char* addr) {
     const uint2 a = *reinterpret_cast<const uint2*>(addr);
@@ -57,13 +50,13 @@ __device__ __forceinline__ short8 lds_frag_b64x2(const char* addr) {
 }
 
 __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams p) {
-    __shared__ char smem[K_SZ + VT_SZ + P_SZ];  // k | vt | p; o aliases k+vt
+    __shared__ char k_lds[KVB * K_ROW];          // [t][d]; reused as o_lds
+    __shared__ char vt_lds[C * VT_ROW];          // [d][t]
+    __shared__ char p_lds[QT * P_ROW];           // [q][t] bf16
+    // split-K: waves ds_add_f32 their partials into ONE tile (4 KB; the
+    // per-wave-slab version cost 16 KB and dropped occupancy to 1 block/CU)
     __shared__ float s_red[KVB][QT];
     __shared__ float m_lds[QT], l_lds[QT], corr_lds[QT];
-    char* k_lds = smem;
-    char* vt_lds = smem + K_SZ;
-    char* p_lds = smem + K_SZ + VT_SZ;
-    static_assert(QT * O_ROW <= K_SZ + VT_SZ);
 
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
@@ -81,20 +74,19 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
     const float scale2 = p.scale * 1.44269504088896340736f;
 
     // ---- Q fragments for this wave's d-slice (held in registers) ----------
-    short8 qf[NQT][KS];
-#pragma unroll
-    for (int qt = 0; qt < NQT; ++qt) {
-        const int64_t qrow = q0 + qt * 32 + lo < p.L ? q0 + qt * 32 + lo : p.L - 1;
+    short8 qf[KS];
+    {
+        const int64_t qrow = q0 + lo < p.L ? q0 + lo : p.L - 1;
         const uint16_t* qp = qb + qrow * C + wave * DSL;
 #pragma unroll
         for (int ks = 0; ks < KS; ++ks)
-            qf[qt][ks] = *reinterpret_cast<const short8*>(qp + ks * 16 + hi * 8);
+            qf[ks] = *reinterpret_cast<const short8*>(qp + ks * 16 + hi * 8);
     }
     if (tid < QT) {
         m_lds[tid] = -1e30f;
         l_lds[tid] = 0.f;
     }
-    float16v ot[NQT][DT] = {};
+    float16v ot[DT] = {};
 
     const int n_tiles = (int)((p.L + KVB - 1) / KVB);
     for (int tile = 0; tile < n_tiles; ++tile) {
@@ -104,6 +96,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
             reinterpret_cast<float*>(s_red)[c] = 0.f;
         // ---- stage K [t][d] and V^T [d][t] --------------------------------
         {
+            // K: 32 rows x 64 x 16 B chunks = 2048 items / 256 threads
             constexpr int KITEMS = KVB * (C / 8) / (NW * WAVE_SIZE);
 #pragma unroll
             for (int it = 0; it < KITEMS; ++it) {
@@ -117,7 +110,10 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
                 *reinterpret_cast<uint2*>(dst) = uint2{raw.x, raw.y};
                 *reinterpret_cast<uint2*>(dst + 8) = uint2{raw.z, raw.w};
             }
-            // V^T via the DPP butterfly: slots = (token-window, 64-d block)
+            // V^T via the DPP butterfly: each pass covers 8 tokens x 64 d
+            // per wave; 4 waves x VPASS passes x 8 d-blocks... iterate
+            // (token-window, d-block) pairs: windows = KVB/8 = 4,
+            // d-blocks = C/64 = 8 -> 32 slots / 4 waves = 8 per wave.
 #pragma unroll
             for (int sl = 0; sl < (KVB / 8) * (C / 64) / NW; ++sl) {
                 const int slot = sl * NW + wave;
@@ -138,28 +134,26 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
         __syncthreads();
 
         // ---- partial S^T for this wave's d-slice --------------------------
+        float16v s = {};
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int qt = 0; qt < NQT; ++qt) {
-            float16v s = {};
-            __builtin_amdgcn_s_setprio(1);
+        for (int ks = 0; ks < KS; ++ks) {
+            const short8 kf = lds_frag_b64x2(
+                &k_lds[lo * K_ROW + (wave * DSL + ks * 16 + hi * 8) * 2]);
+            s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ks], s, 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        // lane holds S^T rows t = (r&3)+8*(r>>2)+4*hi for q col lo
 #pragma unroll
-            for (int ks = 0; ks < KS; ++ks) {
-                const short8 kf = lds_frag_b64x2(
-                    &k_lds[lo * K_ROW + (wave * DSL + ks * 16 + hi * 8) * 2]);
-                s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[qt][ks], s, 0, 0, 0);
-            }
-            __builtin_amdgcn_s_setprio(0);
-            // lane holds S^T rows t = (r&3)+8*(r>>2)+4*hi for q col qt*32+lo
-#pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                const int t = (r & 3) + 8 * (r >> 2) + 4 * hi;
-                atomicAdd(&s_red[t][qt * 32 + lo], s[r]);
-            }
+        for (int r = 0; r < 16; ++r) {
+            const int t = (r & 3) + 8 * (r >> 2) + 4 * hi;
+            atomicAdd(&s_red[t][lo], s[r]);
         }
         __syncthreads();
 
-        // ---- reduce + shared online softmax -------------------------------
-        // thread -> (q = tid/8, 4 tokens); 8 consecutive threads per q.
+        // ---- split-K reduce + shared online softmax -----------------------
+        // thread -> (q = tid/8, 4 tokens); groups of 8 consecutive threads
+        // share one q.
         {
             const int q = tid / 8;
             const int t0l = (tid % 8) * 4;
@@ -173,6 +167,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
                 sv[j] = acc;
                 tmax = fmaxf(tmax, acc);
             }
+            // per-q max across the 8 threads (3 xor-shuffles)
             tmax = fmaxf(tmax, __shfl_xor(tmax, 1, WAVE_SIZE));
             tmax = fmaxf(tmax, __shfl_xor(tmax, 2, WAVE_SIZE));
             tmax = fmaxf(tmax, __shfl_xor(tmax, 4, WAVE_SIZE));
@@ -190,25 +185,22 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
             tsum += __shfl_xor(tsum, 2, WAVE_SIZE);
             tsum += __shfl_xor(tsum, 4, WAVE_SIZE);
             if ((tid % 8) == 0) {
-                const float corr =
-                    m_old == -1e30f ? 1.f
-                                    : __builtin_amdgcn_exp2f((m_old - m_new) * scale2);
-                corr_lds[q] = corr;
-                l_lds[q] = l_lds[q] * corr + tsum;
+                const float corr = __builtin_amdgcn_exp2f((m_old - m_new) * scale2);
+                corr_lds[q] = m_old == -1e30f ? 1.f : corr;
+                l_lds[q] = l_lds[q] * (m_old == -1e30f ? 1.f : corr) + tsum;
                 m_lds[q] = m_new;
             }
         }
         __syncthreads();
 
         // ---- rescale + PV for this wave's d-slice -------------------------
-#pragma unroll
-        for (int qt = 0; qt < NQT; ++qt) {
-            const float corr = corr_lds[qt * 32 + lo];
+        {
+            const float corr = corr_lds[lo];  // lane's q column
             if (corr != 1.f) {
 #pragma unroll
                 for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
-                    for (int r = 0; r < 16; ++r) ot[qt][dt][r] *= corr;
+                    for (int r = 0; r < 16; ++r) ot[dt][r] *= corr;
             }
             __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -216,13 +208,12 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
 #pragma unroll
                 for (int kt = 0; kt < 2; ++kt) {
                     const int d = wave * DSL + dt * 32 + lo;
-                    const short8 vf =
-                        lds_frag_b64x2(&vt_lds[d * VT_ROW + (kt * 16 + hi * 8) * 2]);
+                    const short8 vf = lds_frag_b64x2(
+                        &vt_lds[d * VT_ROW + (kt * 16 + hi * 8) * 2]);
                     const short8 pf = lds_frag_b64x2(
-                        &p_lds[(qt * 32 + lo) * P_ROW + (kt * 16 + hi * 8) * 2]);
+                        &p_lds[lo * P_ROW + (kt * 16 + hi * 8) * 2]);
                     // A = V^T rows d, B = P^T cols q -> O^T[d][q]
-                    ot[qt][dt] =
-                        __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, ot[qt][dt], 0, 0, 0);
+                    ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, ot[dt], 0, 0, 0);
                 }
             }
             __builtin_amdgcn_s_setprio(0);
@@ -231,20 +222,17 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void vae_attn_kernel(VaeAttnParams 
     }
 
     // ---- epilogue: O^T -> LDS [q][d] -> coalesced [L][512] stores ---------
-    char* o_lds = smem;  // aliases k|vt (QT * O_ROW = 66.6 KB <= 70 KB)
+    char* o_lds = k_lds;  // reuse (32 x O_ROW = 33 KB fits the K buffer)
     {
 #pragma unroll
-        for (int qt = 0; qt < NQT; ++qt) {
-            const float inv = l_lds[qt * 32 + lo] > 0.f ? 1.f / l_lds[qt * 32 + lo] : 0.f;
+        for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
-            for (int dt = 0; dt < DT; ++dt)
-#pragma unroll
-                for (int r = 0; r < 16; ++r) {
-                    const int d = wave * DSL + dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-                    reinterpret_cast<bf16_t*>(&o_lds[(qt * 32 + lo) * O_ROW])[d] =
-                        __float2bfloat16(ot[qt][dt][r] * inv);
-                }
-        }
+            for (int r = 0; r < 16; ++r) {
+                const int d = wave * DSL + dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+                const float inv = l_lds[lo] > 0.f ? 1.f / l_lds[lo] : 0.f;
+                reinterpret_cast<bf16_t*>(&o_lds[lo * O_ROW])[d] =
+                    __float2bfloat16(ot[dt][r] * inv);
+            }
     }
     __syncthreads();
     {
