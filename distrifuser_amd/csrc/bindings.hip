@@ -258,6 +258,37 @@ at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
     return o;
 }
 
+at::Tensor layer_norm(const at::Tensor& x_, const at::Tensor& w_, const at::Tensor& b_,
+                      double eps) {
+    TORCH_CHECK(x_.is_cuda() && x_.scalar_type() == at::kBFloat16);
+    auto x = x_.contiguous();
+    const int C = (int)x.size(-1);
+    TORCH_CHECK(C % 8 == 0 && C <= 2048, "layer_norm kernel: C % 8 == 0 and C <= 2048");
+    auto w = w_.to(at::kBFloat16).contiguous();
+    auto b = b_.to(at::kBFloat16).contiguous();
+    auto y = at::empty_like(x);
+    launch_layer_norm(x.data_ptr(), nullptr, y.data_ptr(), nullptr, w.data_ptr(), b.data_ptr(),
+                      (float)eps, x.numel() / C, C, cur_stream());
+    return y;
+}
+
+std::vector<at::Tensor> add_layer_norm(const at::Tensor& x_, const at::Tensor& res_,
+                                       const at::Tensor& w_, const at::Tensor& b_, double eps) {
+    TORCH_CHECK(x_.is_cuda() && x_.scalar_type() == at::kBFloat16);
+    auto x = x_.contiguous();
+    auto res = res_.contiguous();
+    TORCH_CHECK(res.sizes() == x.sizes() && res.scalar_type() == at::kBFloat16);
+    const int C = (int)x.size(-1);
+    TORCH_CHECK(C % 8 == 0 && C <= 2048, "add_layer_norm kernel: C % 8 == 0 and C <= 2048");
+    auto w = w_.to(at::kBFloat16).contiguous();
+    auto b = b_.to(at::kBFloat16).contiguous();
+    auto y = at::empty_like(x);
+    auto sum = at::empty_like(x);
+    launch_layer_norm(x.data_ptr(), res.data_ptr(), y.data_ptr(), sum.data_ptr(), w.data_ptr(),
+                      b.data_ptr(), (float)eps, x.numel() / C, C, cur_stream());
+    return {sum, y};
+}
+
 at::Tensor vae_attention(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v) {
     TORCH_CHECK(q.is_cuda() && q.dim() == 3 && q.size(-1) == 512 &&
                 q.scalar_type() == at::kBFloat16,
@@ -315,6 +346,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "merge stale peer GN moments with fresh local ones (one launch)");
     m.def("cfg_affine_step", &cfg_affine_step, "fused CFG combine + affine scheduler update");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
+    m.def("layer_norm", &layer_norm, "bf16 fused LayerNorm");
+    m.def("add_layer_norm", &add_layer_norm, "bf16 fused residual-add + LayerNorm -> (sum, y)");
     m.def("vae_attention", &vae_attention, "bf16 single-head d=512 VAE mid attention");
     m.def("conv3x3", &conv3x3,
           "bf16 implicit-GEMM 3x3 conv, stride 1/2, in-place halo rows");

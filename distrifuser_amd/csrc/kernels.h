@@ -65,6 +65,13 @@ struct Conv3x3Params {
 };
 void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream);
 
+// ---- fused (residual +) LayerNorm (bf16, C <= 2048, C % 8 == 0) -------------
+// y = LN(x [+ res]); when res != null and sum_out != null, x+res is also
+// written to sum_out (the transformer residual add fused away).
+void launch_layer_norm(const void* x, const void* res, void* y, void* sum_out,
+                       const void* w, const void* b, float eps, int64_t rows, int C,
+                       hipStream_t stream);
+
 // ---- VAE mid-block attention (bf16, single head, head_dim 512) --------------
 // q/k/v/o: [B][L][512] row-contiguous (sb = batch stride in elements).
 struct VaeAttnParams {

@@ -5,6 +5,8 @@ from __future__ import annotations
 import torch
 from torch import nn
 
+from .. import ops
+
 from .layers import LayerFactory
 
 
@@ -27,11 +29,14 @@ class BasicTransformerBlock(nn.Module):
         self.ff = factory.feed_forward(dim)
 
     def forward(self, x: torch.Tensor, encoder_hidden_states: torch.Tensor) -> torch.Tensor:
-        x = x + self.attn1(self.norm1(x))
-        h = self.norm2(x)
-        x = x + self.attn2(h, encoder_hidden_states)
-        x = x + self.ff(self.norm3(x))
-        return x
+        # residual adds fused into the next LayerNorm (one kernel emits
+        # both the running sum and the normalized view — csrc/layernorm.hip)
+        n1 = ops.layer_norm(x, self.norm1.weight, self.norm1.bias, self.norm1.eps)
+        x, h = ops.add_layer_norm(x, self.attn1(n1), self.norm2.weight, self.norm2.bias,
+                                  self.norm2.eps)
+        x, n3 = ops.add_layer_norm(x, self.attn2(h, encoder_hidden_states),
+                                   self.norm3.weight, self.norm3.bias, self.norm3.eps)
+        return x + self.ff(n3)
 
 
 class Transformer2DModel(nn.Module):

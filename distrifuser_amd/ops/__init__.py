@@ -9,6 +9,7 @@ Set DFA_FORCE_EAGER=1 to force the eager path on GPU (A/B debugging only).
 
 from .conv import NativeConv2d, conv3x3_halo, pack_conv3x3_weight
 from .dispatch import (
+    add_layer_norm,
     flash_attention,
     flash_attention_chunked,
     geglu,
@@ -16,12 +17,15 @@ from .dispatch import (
     group_norm_silu,
     group_norm_stats,
     hip_ext,
-    vae_attention,
     hip_ext_available,
+    layer_norm,
+    vae_attention,
 )
 
 __all__ = [
     "NativeConv2d",
+    "add_layer_norm",
+    "layer_norm",
     "conv3x3_halo",
     "flash_attention",
     "flash_attention_chunked",

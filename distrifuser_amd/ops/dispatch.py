@@ -135,6 +135,20 @@ def group_norm_silu(x, num_groups, weight, bias, eps, silu=True):
     return eager.group_norm_silu(x, num_groups, weight, bias, eps, silu)
 
 
+def layer_norm(x, weight, bias, eps):
+    if _use_hip(x) and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0 and x.shape[-1] <= 2048:
+        return hip_ext().layer_norm(x, weight, bias, eps)
+    return eager.layer_norm(x, weight, bias, eps)
+
+
+def add_layer_norm(x, res, weight, bias, eps):
+    """(x + res, LN(x + res)) in one kernel on GPU."""
+    if _use_hip(x) and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0 and x.shape[-1] <= 2048:
+        s, y = hip_ext().add_layer_norm(x, res, weight, bias, eps)
+        return s, y
+    return eager.add_layer_norm(x, res, weight, bias, eps)
+
+
 def vae_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
     """Single-head d=512 attention, q/k/v [B, L, 512] (VAE mid block)."""
     if _use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] == 512:

@@ -89,6 +89,20 @@ def geglu(hidden: torch.Tensor) -> torch.Tensor:
     return a * F.gelu(b)
 
 
+def layer_norm(x, weight, bias, eps):
+    import torch.nn.functional as _F
+
+    return _F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+def add_layer_norm(x, res, weight, bias, eps):
+    """Fused residual add + LayerNorm: returns (x + res, LN(x + res))."""
+    import torch.nn.functional as _F
+
+    s = x + res
+    return s, _F.layer_norm(s, (s.shape[-1],), weight, bias, eps)
+
+
 def vae_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
     """Single-head attention over [B, L, C] (fp32 softmax)."""
     scale = q.shape[-1] ** -0.5

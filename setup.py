@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "conv.hip"),
         os.path.join(CSRC, "vae_attn.hip"),
+        os.path.join(CSRC, "layernorm.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++20"],

@@ -379,3 +379,40 @@ def test_vae_attention_vs_fp32(b, l):
     ref = eager.vae_attention(q.float(), k.float(), v.float())
     err = (got.float() - ref).abs().max().item()
     assert err <= 0.03, f"max err {err}"
+
+
+# ---- fused (residual +) LayerNorm ------------------------------------------
+
+
+@requires_gpu
+@pytest.mark.parametrize("rows,c", [(64, 320), (100, 640), (33, 1280), (17, 2048)])
+def test_layer_norm_gpu(rows, c):
+    import torch.nn.functional as F
+
+    from distrifuser_amd import ops
+
+    torch.manual_seed(0)
+    x = torch.randn(rows, c, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(c, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(c, device="cuda", dtype=torch.bfloat16)
+    got = ops.hip_ext().layer_norm(x, w, b, 1e-5)
+    ref = F.layer_norm(x.float(), (c,), w.float(), b.float(), 1e-5)
+    assert (got.float() - ref).abs().max().item() < 0.05
+
+
+@requires_gpu
+def test_add_layer_norm_gpu():
+    import torch.nn.functional as F
+
+    from distrifuser_amd import ops
+
+    torch.manual_seed(1)
+    x = torch.randn(50, 1280, device="cuda", dtype=torch.bfloat16)
+    r = torch.randn(50, 1280, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(1280, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(1280, device="cuda", dtype=torch.bfloat16)
+    s, y = ops.hip_ext().add_layer_norm(x, r, w, b, 1e-5)
+    sref = (x.float() + r.float())
+    yref = F.layer_norm(sref, (1280,), w.float(), b.float(), 1e-5)
+    assert (s.float() - sref).abs().max().item() < 0.05
+    assert (y.float() - yref).abs().max().item() < 0.05
