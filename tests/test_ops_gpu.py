@@ -4,7 +4,7 @@
 import pytest
 import torch
 
-pytestmark = pytest.mark.gpu
+pytestmark = [pytest.mark.gpu, pytest.mark.timeout(300)]
 
 if torch.cuda.is_available():
     from distrifuser_amd import ops
