@@ -98,6 +98,23 @@ class NativeConv2d(nn.Conv2d):
                 x, self.weight, self.bias, self.stride[0], top, bot,
                 packed=self.packed_weight(),
             )
+        if (
+            x.is_cuda
+            and self.kernel_size == (1, 1)
+            and self.stride == (1, 1)
+            and self.groups == 1
+            and top is None
+            and bot is None
+        ):
+            # 1x1 conv as a hipBLASLt GEMM: MIOpen's find phase was spending
+            # ~90 s benchmarking naive kernels for these once the 3x3s moved
+            # to the native kernel (rocprof r02 trace).
+            b, c, h, w = x.shape
+            out = torch.matmul(self.weight.view(self.out_channels, c),
+                               x.reshape(b, c, h * w))
+            if self.bias is not None:
+                out = out + self.bias.view(1, -1, 1)
+            return out.view(b, self.out_channels, h, w)
         if top is None and bot is None:
             return super().forward(x)
         return eager.conv3x3_halo(x, self.weight, self.bias, self.stride[0], top, bot)
