@@ -4,8 +4,13 @@ reference's compute_metrics.py: PSNR / LPIPS / FID).
 PSNR and SSIM are computed natively (numpy/scipy). LPIPS and FID require
 pretrained feature networks (AlexNet/InceptionV3) that cannot be downloaded
 in this offline environment — pass --lpips_weights / --fid_weights pointing
-at local checkpoints to enable them, otherwise they are skipped with a note.
-Folders may contain .npy (HWC uint8) or .png files."""
+at local TorchScript modules to enable them, otherwise they are skipped
+with a note. The metric MATH is native: LPIPS = the provided module applied
+to [-1,1] image pairs (reference used torchmetrics' LPIPS the same way,
+/root/reference/scripts/compute_metrics.py:42-79); FID = Frechet distance
+between feature means/covariances (scipy sqrtm), features from the provided
+module applied to [0,1] images. Folders may contain .npy (HWC uint8) or
+.png files."""
 
 import argparse
 import os
@@ -55,6 +60,64 @@ def ssim(a, b):
     )
 
 
+def _to_batch(imgs, keys):
+    import torch
+
+    arrs = [imgs[k].astype(np.float32) for k in keys]
+    x = torch.from_numpy(np.stack(arrs)).permute(0, 3, 1, 2) / 255.0
+    return x
+
+
+def compute_lpips(model_path, imgs0, imgs1, keys, device="cpu"):
+    """Mean LPIPS over pairs; model = TorchScript taking two [-1,1] NCHW
+    batches and returning per-pair distances (or a single scalar)."""
+    import torch
+
+    model = torch.jit.load(model_path, map_location=device).eval()
+    vals = []
+    with torch.no_grad():
+        for k in keys:
+            a = _to_batch(imgs0, [k]).to(device) * 2 - 1
+            b = _to_batch(imgs1, [k]).to(device) * 2 - 1
+            d = model(a, b)
+            vals.append(float(torch.as_tensor(d).reshape(-1).mean()))
+    return float(np.mean(vals))
+
+
+def frechet_distance(mu1, cov1, mu2, cov2):
+    from scipy import linalg
+
+    diff = mu1 - mu2
+    covmean, _ = linalg.sqrtm(cov1 @ cov2, disp=False)
+    if np.iscomplexobj(covmean):
+        covmean = covmean.real
+    return float(diff @ diff + np.trace(cov1) + np.trace(cov2) - 2 * np.trace(covmean))
+
+
+def compute_fid(model_path, imgs0, imgs1, keys, device="cpu"):
+    """FID between the two folders; model = TorchScript feature extractor
+    taking a [0,1] NCHW batch and returning [N, D] features."""
+    import torch
+
+    model = torch.jit.load(model_path, map_location=device).eval()
+
+    def feats(imgs):
+        out = []
+        with torch.no_grad():
+            for k in keys:
+                f = model(_to_batch(imgs, [k]).to(device))
+                out.append(torch.as_tensor(f).reshape(1, -1).cpu().numpy())
+        return np.concatenate(out, axis=0).astype(np.float64)
+
+    f0, f1 = feats(imgs0), feats(imgs1)
+    mu0, mu1 = f0.mean(0), f1.mean(0)
+    cov0 = np.cov(f0, rowvar=False)
+    cov1 = np.cov(f1, rowvar=False)
+    cov0 = np.atleast_2d(cov0)
+    cov1 = np.atleast_2d(cov1)
+    return frechet_distance(mu0, cov0, mu1, cov1)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--input_root0", type=str, required=True)
@@ -88,10 +151,14 @@ def main():
     print(f"SSIM: {np.mean(ssims):.4f}")
     if args.lpips_weights is None:
         print("LPIPS: skipped (no pretrained AlexNet available offline; "
-              "pass --lpips_weights)")
+              "pass --lpips_weights <torchscript>)")
+    else:
+        print(f"LPIPS: {compute_lpips(args.lpips_weights, imgs0, imgs1, keys):.4f}")
     if args.fid_weights is None:
         print("FID: skipped (no pretrained InceptionV3 available offline; "
-              "pass --fid_weights)")
+              "pass --fid_weights <torchscript>)")
+    else:
+        print(f"FID: {compute_fid(args.fid_weights, imgs0, imgs1, keys):.3f}")
 
 
 if __name__ == "__main__":

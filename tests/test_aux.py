@@ -127,3 +127,40 @@ def test_pretrained_requires_real_tokenizer(tmp_path):
     with _pytest.raises(FileNotFoundError, match="vocab.json"):
         DistriSDXLPipeline.from_pretrained(
             cfg, preset="tiny", pretrained_model_name_or_path=str(tmp_path))
+
+
+def test_compute_metrics_lpips_fid_backends(tmp_path):
+    """LPIPS/FID paths run end-to-end given (tiny, random) TorchScript nets."""
+    import subprocess
+    import sys
+
+    import numpy as np
+    import torch
+
+    class TinyLpips(torch.nn.Module):
+        def forward(self, a, b):
+            return ((a - b) ** 2).mean(dim=(1, 2, 3))
+
+    class TinyFeat(torch.nn.Module):
+        def forward(self, x):
+            return x.mean(dim=(2, 3))  # [N, 3] "features"
+
+    lp = tmp_path / "lpips.pt"
+    fe = tmp_path / "feat.pt"
+    torch.jit.script(TinyLpips()).save(str(lp))
+    torch.jit.script(TinyFeat()).save(str(fe))
+    rng = np.random.default_rng(0)
+    for d in ("a", "b"):
+        (tmp_path / d).mkdir()
+        for i in range(3):
+            np.save(tmp_path / d / f"img{i}.npy",
+                    rng.integers(0, 255, (16, 16, 3), dtype=np.uint8))
+    r = subprocess.run(
+        [sys.executable, "scripts/compute_metrics.py",
+         "--input_root0", str(tmp_path / "a"), "--input_root1", str(tmp_path / "b"),
+         "--lpips_weights", str(lp), "--fid_weights", str(fe)],
+        capture_output=True, text=True, cwd=os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))),
+    )
+    assert r.returncode == 0, r.stderr
+    assert "LPIPS:" in r.stdout and "FID:" in r.stdout and "skipped" not in r.stdout
