@@ -258,6 +258,31 @@ at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
     return o;
 }
 
+std::vector<at::Tensor> cfg_dpm_step(const at::Tensor& noise, const at::Tensor& x,
+                                     const c10::optional<at::Tensor>& x0_prev, double g,
+                                     double ca, double cb, double cc, double cx, double ce) {
+    TORCH_CHECK(noise.is_cuda() && noise.size(0) == 2);
+    auto n = noise.contiguous();
+    auto xc = x.contiguous();
+    TORCH_CHECK(xc.numel() * 2 == n.numel());
+    const void* pp = nullptr;
+    at::Tensor p;
+    if (x0_prev.has_value()) {
+        p = x0_prev->contiguous();
+        TORCH_CHECK(p.numel() == xc.numel() && p.scalar_type() == xc.scalar_type());
+        pp = p.data_ptr();
+    }
+    auto out = at::empty_like(xc);
+    auto x0 = at::empty_like(xc);
+    const int64_t total = xc.numel();
+    launch_cfg_dpm_step(n.data_ptr(),
+                        reinterpret_cast<const char*>(n.data_ptr()) + total * n.element_size(),
+                        xc.data_ptr(), pp, out.data_ptr(), x0.data_ptr(), (float)g, (float)ca,
+                        (float)cb, (float)cc, (float)cx, (float)ce, total, dtype_of(xc),
+                        cur_stream());
+    return {out, x0};
+}
+
 at::Tensor layer_norm(const at::Tensor& x_, const at::Tensor& w_, const at::Tensor& b_,
                       double eps) {
     TORCH_CHECK(x_.is_cuda() && x_.scalar_type() == at::kBFloat16);
@@ -346,6 +371,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "merge stale peer GN moments with fresh local ones (one launch)");
     m.def("cfg_affine_step", &cfg_affine_step, "fused CFG combine + affine scheduler update");
     m.def("flash_attention", &flash_attention, "bf16 d64 flash attention (chunked stale KV)");
+    m.def("cfg_dpm_step", &cfg_dpm_step, "fused CFG + DPM-Solver++(2M) update -> (prev, x0)");
     m.def("layer_norm", &layer_norm, "bf16 fused LayerNorm");
     m.def("add_layer_norm", &add_layer_norm, "bf16 fused residual-add + LayerNorm -> (sum, y)");
     m.def("vae_attention", &vae_attention, "bf16 single-head d=512 VAE mid attention");

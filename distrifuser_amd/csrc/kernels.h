@@ -65,6 +65,12 @@ struct Conv3x3Params {
 };
 void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream);
 
+// ---- fused CFG + DPM-Solver++(2M) step --------------------------------------
+// out = ca*x + cb*eps + cc*x0_prev (x0_prev may be null); x0_out = cx*x + ce*eps
+void launch_cfg_dpm_step(const void* nu, const void* nc, const void* x, const void* x0_prev,
+                         void* out, void* x0_out, float g, float ca, float cb, float cc,
+                         float cx, float ce, int64_t total, int dtype, hipStream_t stream);
+
 // ---- fused (residual +) LayerNorm (bf16, C <= 2048, C % 8 == 0) -------------
 // y = LN(x [+ res]); when res != null and sum_out != null, x+res is also
 // written to sum_out (the transformer residual add fused away).

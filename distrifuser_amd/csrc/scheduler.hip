@@ -81,3 +81,50 @@ void launch_cfg_affine_step(const void* noise_u, const void* noise_c, const void
             break;
     }
 }
+
+namespace {
+
+// DPM-Solver++(2M): x' = ca*x + cb*eps + cc*x0_prev, also emits
+// x0 = cx*x + ce*eps for the next step (x0_prev null on the first step).
+template <typename T>
+__global__ void cfg_dpm_step_kernel(const T* __restrict__ noise_u, const T* __restrict__ noise_c,
+                                    const T* __restrict__ x, const T* __restrict__ x0_prev,
+                                    T* __restrict__ out, T* __restrict__ x0_out, float g,
+                                    float ca, float cb, float cc, float cx, float ce,
+                                    int64_t total) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+        const float nu = to_f32(noise_u[i]);
+        const float eps = nu + g * (to_f32(noise_c[i]) - nu);
+        const float xf = to_f32(x[i]);
+        float prev = ca * xf + cb * eps;
+        if (x0_prev) prev += cc * to_f32(x0_prev[i]);
+        out[i] = from_f32<T>(prev);
+        x0_out[i] = from_f32<T>(cx * xf + ce * eps);
+    }
+}
+
+}  // namespace
+
+void launch_cfg_dpm_step(const void* nu, const void* nc, const void* x, const void* x0_prev,
+                         void* out, void* x0_out, float g, float ca, float cb, float cc,
+                         float cx, float ce, int64_t total, int dtype, hipStream_t stream) {
+    const int block = 256;
+    const int grid = (int)((total + block - 1) / block < 4096 ? (total + block - 1) / block : 4096);
+    switch (dtype) {
+        case DFA_BF16:
+            cfg_dpm_step_kernel<bf16_t><<<grid, block, 0, stream>>>(
+                (const bf16_t*)nu, (const bf16_t*)nc, (const bf16_t*)x, (const bf16_t*)x0_prev,
+                (bf16_t*)out, (bf16_t*)x0_out, g, ca, cb, cc, cx, ce, total);
+            break;
+        case DFA_F16:
+            cfg_dpm_step_kernel<f16_t><<<grid, block, 0, stream>>>(
+                (const f16_t*)nu, (const f16_t*)nc, (const f16_t*)x, (const f16_t*)x0_prev,
+                (f16_t*)out, (f16_t*)x0_out, g, ca, cb, cc, cx, ce, total);
+            break;
+        default:
+            cfg_dpm_step_kernel<float><<<grid, block, 0, stream>>>(
+                (const float*)nu, (const float*)nc, (const float*)x, (const float*)x0_prev,
+                (float*)out, (float*)x0_out, g, ca, cb, cc, cx, ce, total);
+    }
+}

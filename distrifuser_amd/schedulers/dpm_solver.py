@@ -62,3 +62,52 @@ class DPMSolverMultistepScheduler(SchedulerBase):
         self._t_prev = t
         self._step_index += 1
         return prev.to(sample.dtype)
+
+    def guided_step(self, noise: torch.Tensor, timestep, sample: torch.Tensor,
+                    guidance_scale: float) -> torch.Tensor:
+        """CFG combine + DPM-Solver++(2M) update in ONE fused kernel on GPU.
+
+        The 2M update is affine in (x, eps, x0_prev):
+            prev = ca*x + cb*eps + cc*x0_prev,   x0 = cx*x + ce*eps
+        with C = a_p*(exp(-h)-1), C' = C*(1 + 0.5/r0) (order 2) or C (order 1):
+            ca = s_p/s_t - C'/a_t,  cb = C'*s_t/a_t,  cc = 0.5*C/r0.
+        """
+        import os
+
+        if not (noise.is_cuda and os.environ.get("DFA_FORCE_EAGER", "0") != "1"):
+            nu, nc = noise.float().chunk(2)
+            eps = nu + guidance_scale * (nc - nu)
+            return self.step(eps, timestep, sample)
+
+        from ..ops.dispatch import hip_ext
+
+        t = int(timestep)
+        t_prev = self._prev_timestep(t)
+        a_t, s_t, l_t = (float(self.alpha_t[t]), float(self.sigma_t[t]),
+                         float(self.lambda_t[t]))
+        a_p, s_p, l_p = (float(self.alpha_t[t_prev]), float(self.sigma_t[t_prev]),
+                         float(self.lambda_t[t_prev]))
+        import math
+
+        h = l_p - l_t
+        C = a_p * (math.exp(-h) - 1.0)
+        first = (self._x0_prev is None or self.solver_order == 1
+                 or self._step_index == len(self.timesteps) - 1)
+        if first:
+            cprime, cc = C, 0.0
+        else:
+            l_pp = float(self.lambda_t[self._t_prev])
+            r0 = (l_t - l_pp) / h
+            cprime = C * (1.0 + 0.5 / r0)
+            cc = 0.5 * C / r0
+        ca = s_p / s_t - cprime / a_t
+        cb = cprime * s_t / a_t
+        cx = 1.0 / a_t
+        ce = -s_t / a_t
+        prev, x0 = hip_ext().cfg_dpm_step(
+            noise, sample, None if first else self._x0_prev, guidance_scale,
+            ca, cb, cc, cx, ce)
+        self._x0_prev = x0
+        self._t_prev = t
+        self._step_index += 1
+        return prev
