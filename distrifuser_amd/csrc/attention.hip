@@ -184,9 +184,11 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
         }
         __syncthreads();
 
-#pragma unroll
-        for (int st = 0; st < KVB / 32; ++st) {  // 32-token sub-tiles
-            // ---- S^T[kv32][q32] = K_sub x Q^T ----
+        // T15 att[2] double-pipeline (guide): while subtile st's softmax/PV
+        // runs on the VALU, subtile st+1's QK^T fills the OTHER score tile on
+        // the MFMA pipe (separate pipes -> free overlap). Static two-state
+        // ping-pong via full unroll (rule #20: no dynamic indexing).
+        auto qk_tile = [&](int st) {
             float16v s = {};
             __builtin_amdgcn_s_setprio(1);  // favor the MFMA wave (guide T5)
 #pragma unroll
@@ -200,6 +202,14 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
                 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[ks], s, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
+            return s;
+        };
+        float16v s_cur = qk_tile(0);
+#pragma unroll
+        for (int st = 0; st < KVB / 32; ++st) {  // 32-token sub-tiles
+            float16v s_nxt;
+            if (st + 1 < KVB / 32) s_nxt = qk_tile(st + 1);
+            const float16v s = s_cur;
             // lane holds S^T rows crow(r) = (r&3)+8*(r>>2)+4*hi for q col lo
             float tm = -1e30f;
             float pv[16];
@@ -276,6 +286,7 @@ __global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnPar
                 }
             }
             __builtin_amdgcn_s_setprio(0);
+            s_cur = s_nxt;
         }
         __syncthreads();
     }
