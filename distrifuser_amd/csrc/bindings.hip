@@ -269,11 +269,12 @@ std::vector<at::Tensor> cfg_dpm_step(const at::Tensor& noise, const at::Tensor& 
     at::Tensor p;
     if (x0_prev.has_value()) {
         p = x0_prev->contiguous();
-        TORCH_CHECK(p.numel() == xc.numel() && p.scalar_type() == xc.scalar_type());
+        TORCH_CHECK(p.numel() == xc.numel() && p.scalar_type() == at::kFloat,
+                    "x0_prev must be the fp32 state from the previous cfg_dpm_step");
         pp = p.data_ptr();
     }
     auto out = at::empty_like(xc);
-    auto x0 = at::empty_like(xc);
+    auto x0 = at::empty_like(xc, xc.options().dtype(at::kFloat));
     const int64_t total = xc.numel();
     launch_cfg_dpm_step(n.data_ptr(),
                         reinterpret_cast<const char*>(n.data_ptr()) + total * n.element_size(),

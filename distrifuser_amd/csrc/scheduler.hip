@@ -86,10 +86,12 @@ namespace {
 
 // DPM-Solver++(2M): x' = ca*x + cb*eps + cc*x0_prev, also emits
 // x0 = cx*x + ce*eps for the next step (x0_prev null on the first step).
+// x0 state kept in fp32: it reaches ~1/alpha_t * x magnitudes early in the
+// trajectory, where bf16 granularity visibly perturbs the 2M correction.
 template <typename T>
 __global__ void cfg_dpm_step_kernel(const T* __restrict__ noise_u, const T* __restrict__ noise_c,
-                                    const T* __restrict__ x, const T* __restrict__ x0_prev,
-                                    T* __restrict__ out, T* __restrict__ x0_out, float g,
+                                    const T* __restrict__ x, const float* __restrict__ x0_prev,
+                                    T* __restrict__ out, float* __restrict__ x0_out, float g,
                                     float ca, float cb, float cc, float cx, float ce,
                                     int64_t total) {
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -98,9 +100,9 @@ __global__ void cfg_dpm_step_kernel(const T* __restrict__ noise_u, const T* __re
         const float eps = nu + g * (to_f32(noise_c[i]) - nu);
         const float xf = to_f32(x[i]);
         float prev = ca * xf + cb * eps;
-        if (x0_prev) prev += cc * to_f32(x0_prev[i]);
+        if (x0_prev) prev += cc * x0_prev[i];
         out[i] = from_f32<T>(prev);
-        x0_out[i] = from_f32<T>(cx * xf + ce * eps);
+        x0_out[i] = cx * xf + ce * eps;
     }
 }
 
@@ -114,13 +116,13 @@ void launch_cfg_dpm_step(const void* nu, const void* nc, const void* x, const vo
     switch (dtype) {
         case DFA_BF16:
             cfg_dpm_step_kernel<bf16_t><<<grid, block, 0, stream>>>(
-                (const bf16_t*)nu, (const bf16_t*)nc, (const bf16_t*)x, (const bf16_t*)x0_prev,
-                (bf16_t*)out, (bf16_t*)x0_out, g, ca, cb, cc, cx, ce, total);
+                (const bf16_t*)nu, (const bf16_t*)nc, (const bf16_t*)x, (const float*)x0_prev,
+                (bf16_t*)out, (float*)x0_out, g, ca, cb, cc, cx, ce, total);
             break;
         case DFA_F16:
             cfg_dpm_step_kernel<f16_t><<<grid, block, 0, stream>>>(
-                (const f16_t*)nu, (const f16_t*)nc, (const f16_t*)x, (const f16_t*)x0_prev,
-                (f16_t*)out, (f16_t*)x0_out, g, ca, cb, cc, cx, ce, total);
+                (const f16_t*)nu, (const f16_t*)nc, (const f16_t*)x, (const float*)x0_prev,
+                (f16_t*)out, (float*)x0_out, g, ca, cb, cc, cx, ce, total);
             break;
         default:
             cfg_dpm_step_kernel<float><<<grid, block, 0, stream>>>(
