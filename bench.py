@@ -147,7 +147,7 @@ def main() -> None:
             "dtype": "bf16" if use_cuda else "fp32",
             "data": "synthetic prompts, random-init SDXL weights (no network)",
             "config": {
-                "model": "sdxl-base (random init)",
+                "model": ("sdxl-base (random init)" if args.preset == "sdxl" else f"{args.preset} preset (random init)"),
                 "global_batch": 1,
                 "image": f"{args.height}x{args.width}",
                 "seq_len": (args.height // 8) * (args.width // 8),
