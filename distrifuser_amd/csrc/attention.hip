@@ -66,14 +66,8 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float a, float b) {
     return r;
 }
 
-// Second launch_bounds arg: at DPAD 64 the kernel sits at 133 VGPRs — just
-// over the 128 boundary that would allow 4 waves/SIMD. 8-wave blocks at 3
-// waves/SIMD mean only ONE block fits a CU (12 waves < 2x8), so staging and
-// barriers have no cross-block cover (PMC: 41% SQ_WAIT). Forcing the
-// allocator to 128 doubles co-resident blocks.
 template <int NW, int DPAD, int KVB, bool MASK, bool DEFER>
-__global__ __launch_bounds__(NW * WAVE_SIZE, DPAD == 64 ? 4 : 1) void flash_attn_kernel(
-    FlashAttnParams p) {
+__global__ __launch_bounds__(NW * WAVE_SIZE) void flash_attn_kernel(FlashAttnParams p) {
     constexpr int QBLK = QW * NW;
     constexpr int VT_ROW = KVB * 2;         // V^T LDS row bytes [d][t]
     // K rows padded by 8 B: row stride DPAD*2+8 gives gcd(stride/4, 32) = 2,
