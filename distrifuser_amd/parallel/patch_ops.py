@@ -87,6 +87,16 @@ class PatchConv2d(nn.Module):
 
         output_h = h // stride // n
         idx = cfg.split_idx()
+        if self.conv.kernel_size[0] == 3 and padding == 1:
+            # Native path: the rank's input band is a row-slice VIEW of the
+            # full latent and the receptive-field overlap rows are the
+            # kernel's top/bot halo pointers — no F.pad materialization.
+            h0 = output_h * idx * stride
+            h1 = output_h * (idx + 1) * stride
+            band = x[:, :, h0:h1, :]
+            top = x[:, :, h0 - 1 : h0, :] if h0 > 0 else None
+            bot = x[:, :, h1 : h1 + 1, :] if h1 < h else None
+            return self.conv(band, top=top, bot=bot)
         h_begin = output_h * idx * stride - padding
         h_end = output_h * (idx + 1) * stride + padding
         pad = [padding, padding, 0, 0]  # W-left, W-right, H-top, H-bottom
@@ -333,11 +343,27 @@ class PatchSelfAttention(nn.Module):
         self._idx: int | None = None
         self._buffer_view: torch.Tensor | None = None  # [n, B, L, 2*inner]
         self._buffer_list: list[torch.Tensor] | None = None
+        self._wqkv: torch.Tensor | None = None  # cached cat(to_q.w, to_kv.w)
+        self._wqkv_key = None
 
     def reset(self) -> None:
         self._idx = None
         self._buffer_view = None
         self._buffer_list = None
+
+    def _qkv(self, x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+        """ONE fused GEMM for q and kv (the two projections share the input;
+        the kv half is copied into the comm slot regardless, and the flash
+        kernel takes the strided q/kv views directly)."""
+        wq, wkv = self.to_q.weight, self.to_kv.weight
+        key = (wq._version, wkv._version, wq.data_ptr(), wq.dtype)
+        if self._wqkv is None or self._wqkv_key != key:
+            with torch.no_grad():
+                self._wqkv = torch.cat([wq.detach(), wkv.detach()], dim=0).contiguous()
+            self._wqkv_key = key
+        qkv = F.linear(x, self._wqkv)
+        inner = self.to_q.out_features
+        return qkv[..., :inner], qkv[..., inner:]
 
     def _attention(self, q: torch.Tensor, kv: torch.Tensor) -> torch.Tensor:
         """q: [B, Lq, inner]; kv: [B, Lkv, 2*inner] (may be strided)."""
@@ -354,8 +380,7 @@ class PatchSelfAttention(nn.Module):
         state = self.state
         cfg = state.config
         b, l, _ = x.shape
-        q = self.to_q(x)
-        kv = self.to_kv(x)
+        q, kv = self._qkv(x)
 
         if not _is_patch_parallel(state):
             out = self._attention(q, kv)
