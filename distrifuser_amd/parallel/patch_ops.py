@@ -355,6 +355,8 @@ class PatchSelfAttention(nn.Module):
         """ONE fused GEMM for q and kv (the two projections share the input;
         the kv half is copied into the comm slot regardless, and the flash
         kernel takes the strided q/kv views directly)."""
+        if os.environ.get("DFA_NO_QKV_FUSE", "0") == "1":
+            return self.to_q(x), self.to_kv(x)
         wq, wkv = self.to_q.weight, self.to_kv.weight
         key = (wq._version, wkv._version, wq.data_ptr(), wq.dtype)
         if self._wqkv is None or self._wqkv_key != key:
