@@ -440,3 +440,24 @@ def test_cfg_dpm_step_gpu_matches_eager():
         x_b = b.guided_step(noise, int(t), x_b, g)  # eager compose + step
         del os.environ["DFA_FORCE_EAGER"]
         assert (x_a.float() - x_b.float()).abs().max().item() < 0.05, int(t)
+
+
+@requires_gpu
+def test_conv3x3_row_band_views():
+    """conv_in sliced path: interior is a row-slice VIEW of the full latent
+    and the halos are adjacent-row views — must equal the full conv's band."""
+    import torch.nn.functional as F
+
+    from distrifuser_amd.ops import NativeConv2d
+
+    torch.manual_seed(5)
+    full = torch.randn(2, 4, 64, 96, device="cuda", dtype=torch.bfloat16)
+    m = NativeConv2d(4, 320, 3, padding=1).to("cuda", torch.bfloat16)
+    ref = F.conv2d(full.float(), m.weight.float(), m.bias.float(), padding=1)
+    for h0, h1 in ((0, 16), (16, 48), (48, 64)):
+        band = full[:, :, h0:h1]
+        top = full[:, :, h0 - 1 : h0] if h0 > 0 else None
+        bot = full[:, :, h1 : h1 + 1] if h1 < 64 else None
+        got = m(band, top=top, bot=bot)
+        err = (got.float() - ref[:, :, h0:h1]).abs().max().item()
+        assert err <= 0.02 * max(ref.abs().max().item(), 1.0), (h0, h1, err)
