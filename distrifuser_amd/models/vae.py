@@ -160,11 +160,25 @@ class VAEDecoder(nn.Module):
         self.conv_norm_out = nn.GroupNorm(groups, ch[0], eps=1e-6)
         self.conv_out = NativeConv2d(ch[0], config.out_channels, 3, padding=1)
 
+    # -- tiled decode (diffusers AutoencoderKL.enable_tiling parity) --------
+    # The mid-block attention is O(L^2) in latent tokens: a full 3840^2
+    # decode attends over 230k tokens (~109 TFLOP in that one layer). Tiling
+    # decodes overlapping latent tiles independently and linearly blends the
+    # seams — the standard high-resolution VAE path.
+    use_tiling: bool = False
+    tile_latent_size: int = 96
+    tile_overlap: int = 16  # latent pixels blended between neighbouring tiles
+
+    def enable_tiling(self, tile_latent_size: int = 96, tile_overlap: int = 16) -> None:
+        self.use_tiling = True
+        self.tile_latent_size = tile_latent_size
+        self.tile_overlap = tile_overlap
+
+    def disable_tiling(self) -> None:
+        self.use_tiling = False
+
     @torch.no_grad()
-    def decode(self, latents: torch.Tensor) -> torch.Tensor:
-        """latents (scaled) -> images in [-1, 1]."""
-        z = latents / self.config.scaling_factor
-        z = self.post_quant_conv(z)
+    def _decode_one(self, z: torch.Tensor) -> torch.Tensor:
         x = self.conv_in(z)
         x = self.mid_resnet_1(x)
         x = self.mid_attn(x)
@@ -174,5 +188,58 @@ class VAEDecoder(nn.Module):
         x = ops.group_norm_silu(x, self.conv_norm_out.num_groups, self.conv_norm_out.weight,
                                 self.conv_norm_out.bias, self.conv_norm_out.eps)
         return self.conv_out(x)
+
+    @torch.no_grad()
+    def _decode_tiled(self, z: torch.Tensor) -> torch.Tensor:
+        ts, ov = self.tile_latent_size, self.tile_overlap
+        stride = ts - ov
+        b, _, h, w = z.shape
+        sf = 8  # spatial upscale of the decoder
+        rows = []
+        for y0 in range(0, max(h - ov, 1), stride):
+            row = []
+            for x0 in range(0, max(w - ov, 1), stride):
+                tile = z[:, :, y0 : y0 + ts, x0 : x0 + ts]
+                row.append(self._decode_one(tile))
+            rows.append(row)
+
+        def blend_v(a, bt, k):
+            k = min(k, a.shape[2], bt.shape[2])
+            w_ = torch.linspace(0, 1, k, device=a.device, dtype=a.dtype).view(1, 1, k, 1)
+            bt[:, :, :k] = a[:, :, a.shape[2] - k :] * (1 - w_) + bt[:, :, :k] * w_
+            return bt
+
+        def blend_h(a, bt, k):
+            k = min(k, a.shape[3], bt.shape[3])
+            w_ = torch.linspace(0, 1, k, device=a.device, dtype=a.dtype).view(1, 1, 1, k)
+            bt[:, :, :, :k] = a[:, :, :, a.shape[3] - k :] * (1 - w_) + bt[:, :, :, :k] * w_
+            return bt
+
+        ov_px = ov * sf
+        st_px = stride * sf
+        out_rows = []
+        for i, row in enumerate(rows):
+            parts = []
+            for j, tile in enumerate(row):
+                if i > 0:
+                    tile = blend_v(rows[i - 1][j], tile, ov_px)
+                if j > 0:
+                    tile = blend_h(row[j - 1], tile, ov_px)
+                keep_w = st_px if j < len(row) - 1 else tile.shape[3]
+                parts.append(tile[:, :, :st_px, :keep_w] if i < len(rows) - 1
+                             else tile[:, :, :, :keep_w])
+            out_rows.append(torch.cat(parts, dim=3))
+        out = torch.cat(out_rows, dim=2)
+        return out[:, :, : h * sf, : w * sf]
+
+    @torch.no_grad()
+    def decode(self, latents: torch.Tensor) -> torch.Tensor:
+        """latents (scaled) -> images in [-1, 1]."""
+        z = latents / self.config.scaling_factor
+        z = self.post_quant_conv(z)
+        if self.use_tiling and (z.shape[2] > self.tile_latent_size
+                                or z.shape[3] > self.tile_latent_size):
+            return self._decode_tiled(z)
+        return self._decode_one(z)
 
     forward = decode
