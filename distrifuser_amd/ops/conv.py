@@ -80,7 +80,7 @@ class NativeConv2d(nn.Conv2d):
             and self.stride[0] in (1, 2)
             and self.dilation == (1, 1)
             and self.groups == 1
-            and _hip_conv_ok(x, self.stride[0])
+            and x.dtype == torch.bfloat16
         )
 
     def packed_weight(self) -> torch.Tensor:
@@ -94,6 +94,11 @@ class NativeConv2d(nn.Conv2d):
 
     def forward(self, x: torch.Tensor, top=None, bot=None) -> torch.Tensor:
         if self._native_eligible(x):
+            if not _hip_conv_ok(x, self.stride[0]):
+                # e.g. a tiled-decode row-slice view: one contiguous copy is
+                # far cheaper than MIOpen's fallback (a naive-conv trial was
+                # 595 ms/tile in the tiled-VAE trace)
+                x = x.contiguous()
             return conv3x3_halo(
                 x, self.weight, self.bias, self.stride[0], top, bot,
                 packed=self.packed_weight(),
