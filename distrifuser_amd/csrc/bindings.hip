@@ -204,7 +204,8 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
 
 at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
                    const c10::optional<at::Tensor>& bias, int64_t cout, int64_t stride,
-                   const c10::optional<at::Tensor>& top, const c10::optional<at::Tensor>& bot) {
+                   const c10::optional<at::Tensor>& top, const c10::optional<at::Tensor>& bot,
+                   const c10::optional<at::Tensor>& residual) {
     TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kBFloat16,
                 "x must be CUDA bf16 [B,Cin,H,W]");
     TORCH_CHECK(x.stride(3) == 1 && x.stride(2) == x.size(3), "x rows must be contiguous");
@@ -251,6 +252,14 @@ at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
     };
     set_halo(top, p.top, p.t_sb, p.t_sc);
     set_halo(bot, p.bot, p.b_sb, p.b_sc);
+    at::Tensor res_c;
+    if (residual.has_value()) {
+        res_c = residual->contiguous();
+        TORCH_CHECK(res_c.scalar_type() == at::kBFloat16 &&
+                    res_c.sizes() == (at::IntArrayRef{(long)B, (long)cout, (long)Ho, (long)Wo}),
+                    "residual must be bf16 [B,Cout,Ho,Wo]");
+        p.residual = reinterpret_cast<const uint16_t*>(res_c.data_ptr());
+    }
 
     auto o = at::empty({(long)B, (long)cout, (long)Ho, (long)Wo}, x.options());
     p.o = reinterpret_cast<uint16_t*>(o.data_ptr());

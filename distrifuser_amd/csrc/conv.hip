@@ -359,14 +359,16 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
     }
 
 
-    // ---- epilogue: O[b][cout][y][x] = acc + bias ---------------------------
+    // ---- epilogue: O[b][cout][y][x] = acc + bias (+ residual) --------------
     const int y = yb0 + wy;
     if (y >= p.Ho) return;
 #pragma unroll
     for (int px = 0; px < PW; ++px) {
         const int x = xb0 + (wx * PW + px) * 32 + lo;
         if (x >= p.Wo) continue;
-        uint16_t* obase = p.o + (int64_t)b * p.Cout * p.Ho * p.Wo + (int64_t)y * p.Wo + x;
+        const int64_t pix_off = (int64_t)b * p.Cout * p.Ho * p.Wo + (int64_t)y * p.Wo + x;
+        uint16_t* obase = p.o + pix_off;
+        const uint16_t* rbase = p.residual ? p.residual + pix_off : nullptr;
 #pragma unroll
         for (int ct2 = 0; ct2 < NCT; ++ct2) {
             const int ct = cb * NCT + ct2;
@@ -377,6 +379,9 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 if (cout < p.Cout) {
                     float v = acc[ct2][px][r];
                     if (p.bias) v += to_f32(reinterpret_cast<const bf16_t*>(p.bias)[cout]);
+                    if (rbase)
+                        v += to_f32(reinterpret_cast<const bf16_t*>(
+                            rbase)[(int64_t)cout * p.Ho * p.Wo]);
                     obase[(int64_t)cout * p.Ho * p.Wo] =
                         __builtin_bit_cast(uint16_t, __float2bfloat16(v));
                 }

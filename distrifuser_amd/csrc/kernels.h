@@ -55,6 +55,7 @@ struct Conv3x3Params {
     const uint16_t* bot;
     const uint16_t* wp;
     const uint16_t* bias;
+    const uint16_t* residual;  // optional [B][Cout][Ho][Wo]: o += residual
     uint16_t* o;
     int B, Cin, Cout, H, W, Ho, Wo;
     int KS, CT;

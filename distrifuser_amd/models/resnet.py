@@ -93,7 +93,10 @@ class ResnetBlock2D(nn.Module):
                 dist.all_reduce(h, op=dist.ReduceOp.SUM, group=self.state.config.batch_group)
             h = h + self.conv2_bias.view(1, -1, 1, 1)
         else:
-            h = self.conv2(self.norm2(h))
+            # shortcut add fused into the conv2 epilogue (HIP kernel
+            # residual pointer; eager paths fall back to out + residual)
+            res = self.conv_shortcut(x) if self.conv_shortcut is not None else x
+            return self.conv2(self.norm2(h), residual=res)
         if self.conv_shortcut is not None:
             x = self.conv_shortcut(x)
         return x + h

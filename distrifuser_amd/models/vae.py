@@ -56,11 +56,10 @@ class VAEResnetBlock(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         h = self.conv1(ops.group_norm_silu(x, self.norm1.num_groups, self.norm1.weight,
                                            self.norm1.bias, self.norm1.eps))
-        h = self.conv2(ops.group_norm_silu(h, self.norm2.num_groups, self.norm2.weight,
-                                           self.norm2.bias, self.norm2.eps))
-        if self.conv_shortcut is not None:
-            x = self.conv_shortcut(x)
-        return x + h
+        res = self.conv_shortcut(x) if self.conv_shortcut is not None else x
+        return self.conv2(ops.group_norm_silu(h, self.norm2.num_groups, self.norm2.weight,
+                                              self.norm2.bias, self.norm2.eps),
+                          residual=res)
 
 
 def _chunked_single_head_attention(

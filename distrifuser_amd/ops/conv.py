@@ -46,16 +46,18 @@ def _hip_conv_ok(x: torch.Tensor, stride: int) -> bool:
     )
 
 
-def conv3x3_halo(x, weight, bias, stride=1, top=None, bot=None, packed=None):
+def conv3x3_halo(x, weight, bias, stride=1, top=None, bot=None, packed=None, residual=None):
     """Dispatching halo conv. ``packed`` (from :func:`pack_conv3x3_weight`)
-    enables the HIP path; eager falls back to the cat-based oracle."""
+    enables the HIP path; eager falls back to the cat-based oracle.
+    ``residual`` is added in the kernel epilogue (ResBlock shortcut)."""
     if _use_hip(x) and packed is not None and _hip_conv_ok(x, stride):
         cout = weight.shape[0]
         t = top.reshape(top.shape[0], top.shape[1], -1) if top is not None else None
         b = bot.reshape(bot.shape[0], bot.shape[1], -1) if bot is not None else None
         if (t is None or t.stride(-1) == 1) and (b is None or b.stride(-1) == 1):
-            return hip_ext().conv3x3(x, packed, bias, cout, stride, t, b)
-    return eager.conv3x3_halo(x, weight, bias, stride, top, bot)
+            return hip_ext().conv3x3(x, packed, bias, cout, stride, t, b, residual)
+    out = eager.conv3x3_halo(x, weight, bias, stride, top, bot)
+    return out if residual is None else out + residual
 
 
 class NativeConv2d(nn.Conv2d):
@@ -92,7 +94,7 @@ class NativeConv2d(nn.Conv2d):
             self._wp_key = key
         return self._wp
 
-    def forward(self, x: torch.Tensor, top=None, bot=None) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, top=None, bot=None, residual=None) -> torch.Tensor:
         if self._native_eligible(x):
             if not _hip_conv_ok(x, self.stride[0]):
                 # e.g. a tiled-decode row-slice view: one contiguous copy is
@@ -101,7 +103,7 @@ class NativeConv2d(nn.Conv2d):
                 x = x.contiguous()
             return conv3x3_halo(
                 x, self.weight, self.bias, self.stride[0], top, bot,
-                packed=self.packed_weight(),
+                packed=self.packed_weight(), residual=residual,
             )
         if (
             x.is_cuda
@@ -121,5 +123,7 @@ class NativeConv2d(nn.Conv2d):
                 out = out + self.bias.view(1, -1, 1)
             return out.view(b, self.out_channels, h, w)
         if top is None and bot is None:
-            return super().forward(x)
-        return eager.conv3x3_halo(x, self.weight, self.bias, self.stride[0], top, bot)
+            out = super().forward(x)
+            return out if residual is None else out + residual
+        out = eager.conv3x3_halo(x, self.weight, self.bias, self.stride[0], top, bot)
+        return out if residual is None else out + residual

@@ -461,3 +461,19 @@ def test_conv3x3_row_band_views():
         got = m(band, top=top, bot=bot)
         err = (got.float() - ref[:, :, h0:h1]).abs().max().item()
         assert err <= 0.02 * max(ref.abs().max().item(), 1.0), (h0, h1, err)
+
+
+@requires_gpu
+def test_conv3x3_residual_epilogue():
+    from distrifuser_amd.ops import conv as conv_ops
+    from distrifuser_amd.ops import eager
+
+    torch.manual_seed(6)
+    x = torch.randn(2, 64, 24, 48, device="cuda", dtype=torch.bfloat16)
+    res = torch.randn(2, 96, 24, 48, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(96, 64, 3, 3, device="cuda", dtype=torch.bfloat16) * 0.04
+    pk = conv_ops.pack_conv3x3_weight(w)
+    got = conv_ops.conv3x3_halo(x, w, None, 1, packed=pk, residual=res)
+    ref = eager.conv3x3_halo(x.float(), w.float(), None, 1) + res.float()
+    err = (got.float() - ref).abs().max().item()
+    assert err <= 0.03 * max(ref.abs().max().item(), 1.0)
