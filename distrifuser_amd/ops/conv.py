@@ -50,6 +50,9 @@ def conv3x3_halo(x, weight, bias, stride=1, top=None, bot=None, packed=None, res
     """Dispatching halo conv. ``packed`` (from :func:`pack_conv3x3_weight`)
     enables the HIP path; eager falls back to the cat-based oracle.
     ``residual`` is added in the kernel epilogue (ResBlock shortcut)."""
+    import os
+    if residual is not None and os.environ.get("DFA_NO_CONV_RESID", "0") == "1":
+        return conv3x3_halo(x, weight, bias, stride, top, bot, packed=packed) + residual
     if _use_hip(x) and packed is not None and _hip_conv_ok(x, stride):
         cout = weight.shape[0]
         t = top.reshape(top.shape[0], top.shape[1], -1) if top is not None else None
