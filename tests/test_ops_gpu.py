@@ -464,10 +464,11 @@ def test_conv3x3_row_band_views():
 
 
 @requires_gpu
-def test_conv3x3_residual_epilogue():
+def test_conv3x3_residual_epilogue(monkeypatch):
     from distrifuser_amd.ops import conv as conv_ops
     from distrifuser_amd.ops import eager
 
+    monkeypatch.setenv("DFA_CONV_RESID", "1")
     torch.manual_seed(6)
     x = torch.randn(2, 64, 24, 48, device="cuda", dtype=torch.bfloat16)
     res = torch.randn(2, 96, 24, 48, device="cuda", dtype=torch.bfloat16)
