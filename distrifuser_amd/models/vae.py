@@ -194,13 +194,25 @@ class VAEDecoder(nn.Module):
         stride = ts - ov
         b, _, h, w = z.shape
         sf = 8  # spatial upscale of the decoder
-        rows = []
-        for y0 in range(0, max(h - ov, 1), stride):
-            row = []
-            for x0 in range(0, max(w - ov, 1), stride):
-                tile = z[:, :, y0 : y0 + ts, x0 : x0 + ts]
-                row.append(self._decode_one(tile))
-            rows.append(row)
+        ys = list(range(0, max(h - ov, 1), stride))
+        xs = list(range(0, max(w - ov, 1), stride))
+        # Full-size tiles decode as ONE batched pass (288 GB HBM3E holds all
+        # activations comfortably); ragged edge tiles decode individually.
+        tiles = {}
+        full = [(y0, x0) for y0 in ys for x0 in xs
+                if y0 + ts <= h and x0 + ts <= w]
+        if len(full) > 1 and b == 1:
+            batch = torch.cat([z[:, :, y0 : y0 + ts, x0 : x0 + ts]
+                               for (y0, x0) in full])
+            dec = self._decode_one(batch)
+            for i, key in enumerate(full):
+                tiles[key] = dec[i : i + 1]
+        for y0 in ys:
+            for x0 in xs:
+                if (y0, x0) not in tiles:
+                    tiles[(y0, x0)] = self._decode_one(
+                        z[:, :, y0 : y0 + ts, x0 : x0 + ts])
+        rows = [[tiles[(y0, x0)] for x0 in xs] for y0 in ys]
 
         def blend_v(a, bt, k):
             k = min(k, a.shape[2], bt.shape[2])
