@@ -62,6 +62,10 @@ def main() -> None:
     use_graphs = use_cuda and not args.no_cuda_graph and (
         world == 1 or args.force_cuda_graph
     )
+    if args.force_cuda_graph and world > 1:
+        print("WARNING: RCCL collectives inside hipGraph capture HANG on this "
+              "stack (docs/DESIGN.md §5) — --force-cuda-graph at world_size>1 "
+              "is a debug probe, not a fast path", flush=True)
     cfg = DistriConfig(
         height=args.height,
         width=args.width,
