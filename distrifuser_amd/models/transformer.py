@@ -28,15 +28,23 @@ class BasicTransformerBlock(nn.Module):
         self.norm3 = nn.LayerNorm(dim)
         self.ff = factory.feed_forward(dim)
 
-    def forward(self, x: torch.Tensor, encoder_hidden_states: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, encoder_hidden_states: torch.Tensor,
+                pending=None):
         # residual adds fused into the next LayerNorm (one kernel emits
-        # both the running sum and the normalized view — csrc/layernorm.hip)
-        n1 = ops.layer_norm(x, self.norm1.weight, self.norm1.bias, self.norm1.eps)
+        # both the running sum and the normalized view — csrc/layernorm.hip).
+        # `pending` is the PREVIOUS block's un-added ff output, folded into
+        # this block's norm1 the same way; the pair (x, ff_out) is returned
+        # so the chain never materializes a standalone add.
+        if pending is None:
+            n1 = ops.layer_norm(x, self.norm1.weight, self.norm1.bias, self.norm1.eps)
+        else:
+            x, n1 = ops.add_layer_norm(x, pending, self.norm1.weight, self.norm1.bias,
+                                       self.norm1.eps)
         x, h = ops.add_layer_norm(x, self.attn1(n1), self.norm2.weight, self.norm2.bias,
                                   self.norm2.eps)
         x, n3 = ops.add_layer_norm(x, self.attn2(h, encoder_hidden_states),
                                    self.norm3.weight, self.norm3.bias, self.norm3.eps)
-        return x + self.ff(n3)
+        return x, self.ff(n3)
 
 
 class Transformer2DModel(nn.Module):
@@ -87,8 +95,10 @@ class Transformer2DModel(nn.Module):
             x = self.proj_in(x)
             inner = x.shape[1]
             x = x.permute(0, 2, 3, 1).reshape(b, h * w, inner)
+        pending = None
         for block in self.transformer_blocks:
-            x = block(x, encoder_hidden_states)
+            x, pending = block(x, encoder_hidden_states, pending)
+        x = x + pending
         if self.use_linear_projection:
             x = self.proj_out(x)
             x = x.reshape(b, h, w, c).permute(0, 3, 1, 2)
