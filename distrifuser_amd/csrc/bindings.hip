@@ -205,7 +205,8 @@ at::Tensor flash_attention(const at::Tensor& q, const at::Tensor& k, const at::T
 at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
                    const c10::optional<at::Tensor>& bias, int64_t cout, int64_t stride,
                    const c10::optional<at::Tensor>& top, const c10::optional<at::Tensor>& bot,
-                   const c10::optional<at::Tensor>& residual) {
+                   const c10::optional<at::Tensor>& residual,
+                   const c10::optional<at::Tensor>& bias2) {
     TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kBFloat16,
                 "x must be CUDA bf16 [B,Cin,H,W]");
     TORCH_CHECK(x.stride(3) == 1 && x.stride(2) == x.size(3), "x rows must be contiguous");
@@ -252,6 +253,12 @@ at::Tensor conv3x3(const at::Tensor& x, const at::Tensor& wp,
     };
     set_halo(top, p.top, p.t_sb, p.t_sc);
     set_halo(bot, p.bot, p.b_sb, p.b_sc);
+    at::Tensor b2_c;
+    if (bias2.has_value()) {
+        b2_c = bias2->to(at::kBFloat16).contiguous();
+        TORCH_CHECK(b2_c.numel() == (int64_t)B * cout, "bias2 must be [B, Cout]");
+        p.bias2 = reinterpret_cast<const uint16_t*>(b2_c.data_ptr());
+    }
     at::Tensor res_c;
     if (residual.has_value()) {
         res_c = residual->contiguous();

@@ -379,6 +379,9 @@ __global__ __launch_bounds__(YB * XW * WAVE_SIZE) void conv3x3_kernel(Conv3x3Par
                 if (cout < p.Cout) {
                     float v = acc[ct2][px][r];
                     if (p.bias) v += to_f32(reinterpret_cast<const bf16_t*>(p.bias)[cout]);
+                    if (p.bias2)
+                        v += to_f32(reinterpret_cast<const bf16_t*>(
+                            p.bias2)[b * p.Cout + cout]);
                     if (rbase)
                         v += to_f32(reinterpret_cast<const bf16_t*>(
                             rbase)[(int64_t)cout * p.Ho * p.Wo]);

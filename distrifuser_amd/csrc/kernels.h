@@ -56,6 +56,7 @@ struct Conv3x3Params {
     const uint16_t* wp;
     const uint16_t* bias;
     const uint16_t* residual;  // optional [B][Cout][Ho][Wo]: o += residual
+    const uint16_t* bias2;     // optional [B][Cout]: o += bias2 (time-emb add)
     uint16_t* o;
     int B, Cin, Cout, H, W, Ho, Wo;
     int KS, CT;

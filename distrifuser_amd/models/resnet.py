@@ -83,9 +83,10 @@ class ResnetBlock2D(nn.Module):
         self.conv2_bias.data.copy_(sd[f"{prefix}conv2.bias"])
 
     def forward(self, x: torch.Tensor, temb: torch.Tensor) -> torch.Tensor:
-        h = self.conv1(self.norm1(x))
-        t = self.time_emb_proj(F.silu(temb))[:, :, None, None]
-        h = h + t
+        # time-embedding add fused into conv1's epilogue (one scalar per
+        # output row, like the conv bias — csrc/conv.hip bias2)
+        t2 = self.time_emb_proj(F.silu(temb))
+        h = self.conv1(self.norm1(x), bias2=t2)
         if self.tp_pair:
             h = F.silu(self.norm2(h))
             h = self.conv2(h)

@@ -46,24 +46,30 @@ def _hip_conv_ok(x: torch.Tensor, stride: int) -> bool:
     )
 
 
-def conv3x3_halo(x, weight, bias, stride=1, top=None, bot=None, packed=None, residual=None):
+def conv3x3_halo(x, weight, bias, stride=1, top=None, bot=None, packed=None, residual=None,
+                 bias2=None):
     """Dispatching halo conv. ``packed`` (from :func:`pack_conv3x3_weight`)
     enables the HIP path; eager falls back to the cat-based oracle.
-    ``residual`` is added in the kernel epilogue (ResBlock shortcut)."""
+    ``residual`` ([B,Cout,Ho,Wo]) and ``bias2`` ([B,Cout], the time-embedding
+    add) are applied in the kernel epilogue."""
     import os
     # In-epilogue residual measured ~0.7% SLOWER end-to-end than a separate
     # add kernel (the epilogue's cout-strided 2 B residual reads are poorly
     # coalesced vs the add kernel's linear pass) — default OFF, opt in with
-    # DFA_CONV_RESID=1. The kernel capability + numerics test stay.
+    # DFA_CONV_RESID=1. bias2 is one scalar per output ROW (like the conv
+    # bias) and is free, so it stays on by default.
     if residual is not None and os.environ.get("DFA_CONV_RESID", "0") != "1":
-        return conv3x3_halo(x, weight, bias, stride, top, bot, packed=packed) + residual
+        out = conv3x3_halo(x, weight, bias, stride, top, bot, packed=packed, bias2=bias2)
+        return out + residual
     if _use_hip(x) and packed is not None and _hip_conv_ok(x, stride):
         cout = weight.shape[0]
         t = top.reshape(top.shape[0], top.shape[1], -1) if top is not None else None
         b = bot.reshape(bot.shape[0], bot.shape[1], -1) if bot is not None else None
         if (t is None or t.stride(-1) == 1) and (b is None or b.stride(-1) == 1):
-            return hip_ext().conv3x3(x, packed, bias, cout, stride, t, b, residual)
+            return hip_ext().conv3x3(x, packed, bias, cout, stride, t, b, residual, bias2)
     out = eager.conv3x3_halo(x, weight, bias, stride, top, bot)
+    if bias2 is not None:
+        out = out + bias2.to(out.dtype)[:, :, None, None]
     return out if residual is None else out + residual
 
 
@@ -101,7 +107,8 @@ class NativeConv2d(nn.Conv2d):
             self._wp_key = key
         return self._wp
 
-    def forward(self, x: torch.Tensor, top=None, bot=None, residual=None) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, top=None, bot=None, residual=None,
+                bias2=None) -> torch.Tensor:
         if self._native_eligible(x):
             if not _hip_conv_ok(x, self.stride[0]):
                 # e.g. a tiled-decode row-slice view: one contiguous copy is
@@ -110,7 +117,7 @@ class NativeConv2d(nn.Conv2d):
                 x = x.contiguous()
             return conv3x3_halo(
                 x, self.weight, self.bias, self.stride[0], top, bot,
-                packed=self.packed_weight(), residual=residual,
+                packed=self.packed_weight(), residual=residual, bias2=bias2,
             )
         if (
             x.is_cuda
@@ -131,6 +138,8 @@ class NativeConv2d(nn.Conv2d):
             return out.view(b, self.out_channels, h, w)
         if top is None and bot is None:
             out = super().forward(x)
-            return out if residual is None else out + residual
-        out = eager.conv3x3_halo(x, self.weight, self.bias, self.stride[0], top, bot)
+        else:
+            out = eager.conv3x3_halo(x, self.weight, self.bias, self.stride[0], top, bot)
+        if bias2 is not None:
+            out = out + bias2.to(out.dtype)[:, :, None, None]
         return out if residual is None else out + residual

@@ -109,7 +109,7 @@ class PatchConv2d(nn.Module):
         sliced = F.pad(x[:, :, h_begin:h_end, :], pad)
         return F.conv2d(sliced, self.conv.weight, self.conv.bias, stride=stride)
 
-    def _halo_forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
+    def _halo_forward(self, x: torch.Tensor, residual=None, bias2=None) -> torch.Tensor:
         cfg = self.state.config
         comm = self.state.comm_manager
         halo = self.conv.padding[0]
@@ -123,9 +123,9 @@ class PatchConv2d(nn.Module):
                         layer_type="conv2d",
                     )
                 # Registration pass: shapes only; halo numerics don't matter yet.
-                return self.conv(x, residual=residual)
+                return self.conv(x, residual=residual, bias2=bias2)
             if self._idx is None:
-                return self.conv(x, residual=residual)
+                return self.conv(x, residual=residual, bias2=bias2)
             self._buffer_list = comm.get_buffer_list(self._idx)
 
         comm.wait(self._idx)
@@ -139,12 +139,12 @@ class PatchConv2d(nn.Module):
         n = cfg.n_device_per_batch
         top = None if split == 0 else self._buffer_list[split - 1][1]
         bot = None if split == n - 1 else self._buffer_list[split + 1][0]
-        out = self._conv_with_halos(x, top, bot, residual)
+        out = self._conv_with_halos(x, top, bot, residual, bias2)
         if not self.state.use_sync_comm and cfg.mode != "no_sync":
             comm.enqueue(self._idx, boundary)
         return out
 
-    def _conv_with_halos(self, x, top, bot, residual=None):
+    def _conv_with_halos(self, x, top, bot, residual=None, bias2=None):
         """Convolve the local band with neighbour halo rows WITHOUT
         materializing cat([halo, x, halo]) (the reference copies the whole
         input per conv, pp/conv2d.py:72-88): run the conv zero-padded on x
@@ -170,21 +170,25 @@ class PatchConv2d(nn.Module):
             padded = torch.cat(parts, dim=2) if len(parts) > 1 else parts[0]
             padded = F.pad(padded, [0, 0, pad_top, pad_bot])
             out = F.conv2d(padded, conv.weight, conv.bias, stride=s, padding=(0, pw))
+            if bias2 is not None:
+                out = out + bias2.to(out.dtype)[:, :, None, None]
             return out if residual is None else out + residual
 
         # NativeConv2d reads the halo rows in place (HIP kernel: the top/bot
         # pointers, SURVEY K4 — no cat, no boundary-row recompute pass)
-        return conv(x, top=top, bot=bot, residual=residual)
+        return conv(x, top=top, bot=bot, residual=residual, bias2=bias2)
 
-    def forward(self, x: torch.Tensor, residual=None) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, residual=None, bias2=None) -> torch.Tensor:
         if not _is_patch_parallel(self.state):
-            return self.conv(x, residual=residual)
+            return self.conv(x, residual=residual, bias2=bias2)
         if self.is_first_layer:
             return self._sliced_forward(x)
         if self.conv.padding[0] == 0:
             out = self.conv(x)  # 1x1: purely local
+            if bias2 is not None:
+                out = out + bias2.to(out.dtype)[:, :, None, None]
             return out if residual is None else out + residual
-        return self._halo_forward(x, residual)
+        return self._halo_forward(x, residual, bias2)
 
 
 class PatchGroupNorm(nn.Module):

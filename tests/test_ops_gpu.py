@@ -478,3 +478,22 @@ def test_conv3x3_residual_epilogue(monkeypatch):
     ref = eager.conv3x3_halo(x.float(), w.float(), None, 1) + res.float()
     err = (got.float() - ref).abs().max().item()
     assert err <= 0.03 * max(ref.abs().max().item(), 1.0)
+
+
+@requires_gpu
+def test_conv3x3_bias2_epilogue():
+    """Per-(batch, channel) bias2 (the fused time-embedding add)."""
+    from distrifuser_amd.ops import conv as conv_ops
+    from distrifuser_amd.ops import eager
+
+    torch.manual_seed(7)
+    x = torch.randn(2, 64, 20, 40, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(96, 64, 3, 3, device="cuda", dtype=torch.bfloat16) * 0.04
+    bias = torch.randn(96, device="cuda", dtype=torch.bfloat16)
+    b2 = torch.randn(2, 96, device="cuda", dtype=torch.bfloat16)
+    pk = conv_ops.pack_conv3x3_weight(w)
+    got = conv_ops.conv3x3_halo(x, w, bias, 1, packed=pk, bias2=b2)
+    ref = eager.conv3x3_halo(x.float(), w.float(), bias.float(), 1) \
+        + b2.float()[:, :, None, None]
+    err = (got.float() - ref).abs().max().item()
+    assert err <= 0.03 * max(ref.abs().max().item(), 1.0)
