@@ -90,3 +90,57 @@ def test_conv3x3_halo_oracle_matches_full_conv():
     x0 = full[:, :, :4]
     got0 = eager.conv3x3_halo(x0, w, b, 1, None, full[:, :, 4:5])
     assert torch.allclose(got0, F.conv2d(full, w, b, padding=1)[:, :, :4], atol=1e-4)
+
+
+def test_transformer_pending_residual_chain():
+    """The fused add+LN residual threading equals the naive block math."""
+    import torch
+
+    from distrifuser_amd.models.layers import LayerFactory
+    from distrifuser_amd.models.transformer import BasicTransformerBlock
+    from distrifuser_amd.parallel.state import ParallelState
+    from distrifuser_amd.utils.config import DistriConfig
+
+    torch.manual_seed(0)
+    cfg = DistriConfig(height=64, width=64, use_cuda_graph=False, device="cpu")
+    fac = LayerFactory(ParallelState(cfg))
+    blocks = [BasicTransformerBlock(32, 2, 16, 16, factory=fac).eval() for _ in range(3)]
+    x = torch.randn(1, 24, 32)
+    ehs = torch.randn(1, 7, 16)
+
+    with torch.no_grad():
+        # naive composition
+        def naive(b, x):
+            import torch.nn.functional as F
+
+            x = x + b.attn1(F.layer_norm(x, (32,), b.norm1.weight, b.norm1.bias, b.norm1.eps))
+            h = F.layer_norm(x, (32,), b.norm2.weight, b.norm2.bias, b.norm2.eps)
+            x = x + b.attn2(h, ehs)
+            return x + b.ff(F.layer_norm(x, (32,), b.norm3.weight, b.norm3.bias, b.norm3.eps))
+
+        ref = x
+        for b in blocks:
+            ref = naive(b, ref)
+
+        got, pending = x, None
+        for b in blocks:
+            got, pending = b(got, ehs, pending)
+        got = got + pending
+    assert torch.allclose(got, ref, atol=1e-5)
+
+
+def test_native_conv2d_bias2_residual_cpu():
+    import torch
+
+    from distrifuser_amd.ops import NativeConv2d
+
+    torch.manual_seed(1)
+    m = NativeConv2d(8, 12, 3, padding=1).eval()
+    x = torch.randn(2, 8, 6, 10)
+    b2 = torch.randn(2, 12)
+    res = torch.randn(2, 12, 6, 10)
+    with torch.no_grad():
+        got = m(x, bias2=b2, residual=res)
+        ref = torch.nn.functional.conv2d(x, m.weight, m.bias, padding=1) \
+            + b2[:, :, None, None] + res
+    assert torch.allclose(got, ref, atol=1e-5)
