@@ -85,8 +85,12 @@ class ResnetBlock2D(nn.Module):
     def forward(self, x: torch.Tensor, temb: torch.Tensor) -> torch.Tensor:
         # time-embedding add fused into conv1's epilogue (one scalar per
         # output row, like the conv bias — csrc/conv.hip bias2)
+        import os
         t2 = self.time_emb_proj(F.silu(temb))
-        h = self.conv1(self.norm1(x), bias2=t2)
+        if os.environ.get("DFA_NO_TEMB_FUSE", "0") == "1":
+            h = self.conv1(self.norm1(x)) + t2[:, :, None, None]
+        else:
+            h = self.conv1(self.norm1(x), bias2=t2)
         if self.tp_pair:
             h = F.silu(self.norm2(h))
             h = self.conv2(h)

@@ -95,10 +95,16 @@ class Transformer2DModel(nn.Module):
             x = self.proj_in(x)
             inner = x.shape[1]
             x = x.permute(0, 2, 3, 1).reshape(b, h * w, inner)
-        pending = None
-        for block in self.transformer_blocks:
-            x, pending = block(x, encoder_hidden_states, pending)
-        x = x + pending
+        import os
+        if os.environ.get("DFA_NO_PENDING_CHAIN", "0") == "1":
+            for block in self.transformer_blocks:
+                x, p = block(x, encoder_hidden_states, None)
+                x = x + p
+        else:
+            pending = None
+            for block in self.transformer_blocks:
+                x, pending = block(x, encoder_hidden_states, pending)
+            x = x + pending
         if self.use_linear_projection:
             x = self.proj_out(x)
             x = x.reshape(b, h, w, c).permute(0, 3, 1, 2)
