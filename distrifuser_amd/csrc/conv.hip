@@ -407,6 +407,20 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
         const int npix = nxb * nyb * p.B;
         const int ncb = (p.CT + NCT - 1) / NCT;
         dim3 grid((unsigned)(8 * ncb * ((npix + 7) / 8)));
+        if (getenv("DFA_CONV_SMALLW") && p.Wo <= 128) {
+            // occupancy experiment for the 120^2 shapes: small tiles, low
+            // VGPR, 2 blocks/CU
+            constexpr int YB2 = 8, XW2 = 1, NCT2 = 2, PW2 = 1;
+            constexpr int XB2 = XW2 * 32 * PW2;
+            const int nxb2 = (p.Wo + XB2 - 1) / XB2;
+            const int nyb2 = (p.Ho + YB2 - 1) / YB2;
+            const int npix2 = nxb2 * nyb2 * p.B;
+            const int ncb2 = (p.CT + NCT2 - 1) / NCT2;
+            dim3 g2((unsigned)(8 * ncb2 * ((npix2 + 7) / 8)));
+            conv3x3_kernel<1, YB2, XW2, 16, NCT2, PW2>
+                <<<g2, dim3(YB2 * XW2 * WAVE_SIZE), 0, stream>>>(p);
+            return;
+        }
         conv3x3_kernel<1, YB, XW, 16, NCT, PW><<<grid, dim3(YB * XW * WAVE_SIZE), 0, stream>>>(p);
     } else {
         // stride 2: 8 waves: 4 output rows x 2 x-waves of 32 px;
