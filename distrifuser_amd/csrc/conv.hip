@@ -407,9 +407,11 @@ void launch_conv3x3(const Conv3x3Params& p, int stride, hipStream_t stream) {
         const int npix = nxb * nyb * p.B;
         const int ncb = (p.CT + NCT - 1) / NCT;
         dim3 grid((unsigned)(8 * ncb * ((npix + 7) / 8)));
-        if (getenv("DFA_CONV_SMALLW") && p.Wo <= 128) {
-            // occupancy experiment for the 120^2 shapes: small tiles, low
-            // VGPR, 2 blocks/CU
+        if (p.Wo <= 128) {
+            // small-W config (SDXL's 120^2 mid/up shapes): 32-px tiles at
+            // 127 VGPRs / 64 KB LDS -> 4 waves/SIMD, 2 blocks/CU; measured
+            // +5% over the wide config at 1280@120^2 (the wide config's
+            // 64-px tiles also waste 6% on the 120->128 x-pad there)
             constexpr int YB2 = 8, XW2 = 1, NCT2 = 2, PW2 = 1;
             constexpr int XB2 = XW2 * 32 * PW2;
             const int nxb2 = (p.Wo + XB2 - 1) / XB2;
