@@ -42,11 +42,6 @@ constexpr int QW = 32;     // q rows per wave
 // covers the SD-family head dims (SDXL/SD2: 64; SD1.5: 40->64, 80->96,
 // 160). The real head_dim rides in FlashAttnParams.Dh; padding lanes carry
 // zeros (exact for QK^T scores and O columns < Dh).
-constexpr int pow2ceil(int v) {
-    int p = 1;
-    while (p < v) p <<= 1;
-    return p;
-}
 
 typedef float float4v_ __attribute__((ext_vector_type(4)));
 typedef float float16v __attribute__((ext_vector_type(16)));

@@ -54,11 +54,6 @@ __device__ __forceinline__ int32x4_ck make_srsrc(const void* base, uint32_t byte
     return r;
 }
 
-template <int ROW_BYTES>
-__device__ __forceinline__ int cv_swz(int row, int byte_off) {
-    static_assert((ROW_BYTES & (ROW_BYTES - 1)) == 0);
-    return byte_off ^ ((row & (ROW_BYTES / 16 - 1)) << 4);
-}
 
 // S: conv stride; (YB rows) x (XW*32*PW x) output tile per block; CIN_T cins
 // per staged LDS tile; each wave owns PW 32-x subtiles x NCT 32-cout tiles
