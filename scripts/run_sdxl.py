@@ -39,6 +39,8 @@ def get_args():
     # Parallelism
     p.add_argument("--no_split_batch", action="store_true")
     p.add_argument("--warmup_steps", type=int, default=4)
+    p.add_argument("--tile_decode", action="store_true",
+                   help="force tiled VAE decode (auto-enabled >= 2048^2)")
     p.add_argument("--sync_mode", type=str, default="corrected_async_gn",
                    choices=["separate_gn", "stale_gn", "corrected_async_gn", "sync_gn",
                             "full_sync", "no_sync"])
@@ -76,6 +78,8 @@ def main():
         cfg, torch_dtype=dtype, scheduler=args.scheduler,
         pretrained_model_name_or_path=args.pretrained, preset=args.preset,
     )
+    if args.tile_decode:
+        pipe.vae.enable_tiling()
 
     def run(output_type):
         g = torch.Generator().manual_seed(args.seed)
